@@ -1,0 +1,158 @@
+"""Checkpoint save/load in the convert2ckpt on-disk layout.
+
+The layout contract (produced by the reference's
+convert2ckpt.py:19-48, consumed via the monkey-patched engine loader at
+trainer_base_ds_mp.py:49-121,284):
+
+    <dir>/latest                                   text: e.g. "global_step001"
+    <dir>/global_stepNNN/
+        layer_{i:02d}-model_00-model_states.pt     state_dict of flat layer i
+        mp_rank_00_model_states.pt                 engine metadata dict
+        engine_state_pp{stage:02d}_dp{dp:02d}.pt   (ours) optimizer/scheduler
+                                                   shards — absent in a
+                                                   converted-from-HF dir
+
+Layer file numbering == flat LayerSpec index (embedding 0, decoder i -> i+1,
+norm L+1, head L+2) — matching convert2ckpt.py:23-36 so converted HF
+checkpoints load per-stage.  Module-only loads (a dir with no engine_state
+files) work natively — the very capability the reference had to monkey-patch
+into DeepSpeed (trainer_base_ds_mp.py:48, README.md:163).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import re
+from pathlib import Path
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+logger = logging.getLogger(__name__)
+
+LAYER_FMT = "layer_{idx:02d}-model_00-model_states.pt"
+MP_RANK_FMT = "mp_rank_{mp:02d}_model_states.pt"
+ENGINE_FMT = "engine_state_pp{stage:02d}_dp{dp:02d}.pt"
+LATEST = "latest"
+
+
+def layer_file(idx: int) -> str:
+    return LAYER_FMT.format(idx=idx)
+
+
+def read_latest(ckpt_dir: str) -> Optional[str]:
+    p = Path(ckpt_dir) / LATEST
+    if not p.exists():
+        return None
+    tag = p.read_text().strip()
+    return tag or None
+
+
+def write_latest(ckpt_dir: str, tag: str) -> None:
+    (Path(ckpt_dir) / LATEST).write_text(tag)
+
+
+def save_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
+                           client_state: Optional[dict] = None) -> str:
+    """All ranks call this collectively.  dp_id==0 ranks write their stage's
+    layer files; every rank writes its engine-state shard; global rank 0
+    writes metadata + latest."""
+    grid = engine.grid
+    if tag is None:
+        tag = f"global_step{engine.global_steps}"
+    step_dir = Path(ckpt_dir) / tag
+    step_dir.mkdir(parents=True, exist_ok=True)
+
+    module = engine.module
+    if grid.dp_id == 0:
+        for local_idx, layer in enumerate(module.layers):
+            gidx = module.global_layer_index(local_idx)
+            sd = {k: v.detach().cpu() for k, v in layer.state_dict().items()}
+            torch.save(sd, step_dir / layer_file(gidx))
+
+    es = engine.state_dict_local()
+    if client_state:
+        es["client_state"] = client_state
+    torch.save(es, step_dir / ENGINE_FMT.format(stage=grid.stage_id, dp=grid.dp_id))
+
+    if grid.rank == 0:
+        meta = {
+            "dp_world_size": grid.dp_degree,
+            "mp_world_size": 1,
+            "num_stages": grid.num_stages,
+            "module": None,
+            "optimizer": None,
+            "global_steps": engine.global_steps,
+            "skipped_steps": engine.skipped_steps,
+            "iteration": engine.global_steps,
+        }
+        torch.save(meta, step_dir / MP_RANK_FMT.format(mp=0))
+        write_latest(ckpt_dir, tag)
+
+    if dist.is_initialized():
+        dist.barrier()
+    return str(step_dir)
+
+
+def load_module_weights(module, ckpt_dir: str, tag: Optional[str] = None,
+                        strict: bool = True, dtype: Optional[torch.dtype] = None) -> str:
+    """Module-only warm start: map this stage's layer files onto local layers.
+    Works on converted-HF dirs (no optimizer state present)."""
+    tag = tag or read_latest(ckpt_dir)
+    if tag is None:
+        raise FileNotFoundError(f"no 'latest' tag in {ckpt_dir}")
+    step_dir = Path(ckpt_dir) / tag
+    for local_idx, layer in enumerate(module.layers):
+        gidx = module.global_layer_index(local_idx)
+        f = step_dir / layer_file(gidx)
+        if not f.exists():
+            raise FileNotFoundError(f"missing layer file {f}")
+        sd = torch.load(f, map_location="cpu", weights_only=True)
+        if dtype is not None:
+            sd = {k: (v.to(dtype) if v.is_floating_point() else v) for k, v in sd.items()}
+        missing, unexpected = layer.load_state_dict(sd, strict=strict)
+        if strict and (missing or unexpected):
+            raise KeyError(f"layer {gidx}: missing={missing} unexpected={unexpected}")
+    # move to module device/dtype handled by caller
+    return str(step_dir)
+
+
+def load_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
+                           load_module_only: bool = False) -> Optional[dict]:
+    """Full resume (module + optimizer + scheduler) or module-only warm start
+    (load_module_only=True — the reference's load path at
+    trainer_base_ds_mp.py:284 with the same flag)."""
+    grid = engine.grid
+    tag = tag or read_latest(ckpt_dir)
+    if tag is None:
+        raise FileNotFoundError(f"no 'latest' tag in {ckpt_dir}")
+    load_module_weights(engine.module, ckpt_dir, tag, dtype=engine.dtype)
+    engine.module.to(engine.device)
+    # refresh masters from (re)loaded params
+    with torch.no_grad():
+        for p, m in zip(engine.optimizer.params, engine.optimizer.masters):
+            m.copy_(p.detach().float())
+
+    client_state = None
+    if not load_module_only:
+        f = Path(ckpt_dir) / tag / ENGINE_FMT.format(stage=grid.stage_id, dp=grid.dp_id)
+        if f.exists():
+            es = torch.load(f, map_location="cpu", weights_only=False)
+            engine.load_state_dict_local(es)
+            client_state = es.get("client_state")
+        else:
+            logger.warning(
+                "no engine state shard at %s — module-only load (converted checkpoint?)", f
+            )
+    if dist.is_initialized():
+        dist.barrier()
+    return client_state
+
+
+def parse_checkpoint_step(path: str) -> int:
+    """checkpoint-500 / global_step500 -> 500 (resume-step parsing the
+    reference does on the dir name, trainer_base_ds_mp.py:452-455)."""
+    m = re.search(r"(?:checkpoint-|global_step)(\d+)", os.path.basename(os.path.normpath(path)))
+    return int(m.group(1)) if m else 0

@@ -1,0 +1,311 @@
+"""PipelineEngine: 1F1B microbatch training over RCCL/xGMI.
+
+The native core that replaces DeepSpeed's PipelineEngine as the reference
+drives it — ``engine.train_batch(data_iter)`` runs ``gradient_accumulation_
+steps`` microbatches through a fill-drain 1F1B schedule, exchanges
+activations/gradients with neighbour stages, all-reduces DP gradients,
+clips, steps AdamW + LR, and returns the mean loss on every rank
+(trainer_base_ds_mp.py:354).
+
+Schedule (per stage ``s`` of ``P``, ``M`` microbatches):
+  warmup   = min(P - 1 - s, M) forwards,
+  steady   = M - warmup 1F1B pairs (forward + oldest-outstanding backward),
+  cooldown = warmup backwards.
+Only the first and last stages touch the data iterator (README.md:64-67 —
+the property that enables the reference's placeholder-dataset trick; here
+middle stages simply do not need a dataset at all).
+
+Precision: bf16 activations/params, fp32 gradient accumulation (hooks into
+the optimizer's flat fp32 buffer), fp32 master AdamW.  fp16 with dynamic
+loss scaling is supported for reference parity (conf/...yaml:137-143).
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from collections import deque
+from typing import Iterator, Optional
+
+import torch
+import torch.distributed as dist
+
+from .config import TrainConfig, torch_dtype
+from .optim import MixedPrecisionAdamW, WarmupDecayLR
+from .p2p import PipeP2P
+from .pipeline_module import PipelineModule
+from .topology import ProcessGrid
+
+logger = logging.getLogger(__name__)
+
+
+class DynamicLossScaler:
+    """fp16 dynamic loss scaling (init 2^12, window 1000, hysteresis 2,
+    min 1 — conf/...yaml:137-143)."""
+
+    def __init__(self, init_scale=2.0**12, scale_window=1000, hysteresis=2, min_scale=1.0):
+        self.scale = init_scale
+        self.scale_window = scale_window
+        self.hysteresis = hysteresis
+        self.min_scale = min_scale
+        self._good_steps = 0
+        self._hyst = hysteresis
+
+    def update(self, found_inf: bool) -> None:
+        if found_inf:
+            self._hyst -= 1
+            if self._hyst <= 0:
+                self.scale = max(self.scale / 2.0, self.min_scale)
+                self._hyst = self.hysteresis
+            self._good_steps = 0
+        else:
+            self._good_steps += 1
+            if self._good_steps >= self.scale_window:
+                self.scale *= 2.0
+                self._good_steps = 0
+
+
+class PipelineEngine:
+    def __init__(self, module: PipelineModule, config: TrainConfig, grid: ProcessGrid,
+                 device: Optional[torch.device] = None):
+        self.module = module
+        self.config = config
+        self.grid = grid
+        self.dtype = torch_dtype(config.dtype)
+        if device is None:
+            device = torch.device("cuda", torch.cuda.current_device()) \
+                if torch.cuda.is_available() else torch.device("cpu")
+        self.device = device
+
+        self.module.to(device)
+        if self.dtype != torch.float32:
+            self.module.to(self.dtype)
+
+        self.micro_batch_size = config.micro_batch_size
+        self.micro_batches = config.gradient_accumulation_steps
+        self.seq_len = config.seq_len
+        hidden = config.model.hidden_size
+        act_shape = (self.micro_batch_size, self.seq_len, hidden)
+        comm_dtype = self.dtype if self.dtype != torch.float32 else torch.float32
+        self.p2p = PipeP2P(grid, act_shape, comm_dtype, device)
+
+        opt_cfg = config.optimizer
+        self.optimizer = MixedPrecisionAdamW(
+            self.module.parameters(),
+            lr=opt_cfg.lr,
+            betas=tuple(opt_cfg.betas),
+            eps=opt_cfg.eps,
+            weight_decay=opt_cfg.weight_decay,
+        )
+        warmup = opt_cfg.warmup_steps or max(1, int(opt_cfg.warmup_proportion * opt_cfg.total_num_steps))
+        self.lr_scheduler = WarmupDecayLR(
+            self.optimizer, warmup, opt_cfg.total_num_steps, opt_cfg.lr
+        )
+        self.loss_scaler = DynamicLossScaler() if config.dtype == "fp16" else None
+        self.global_steps = 0
+        self.skipped_steps = 0
+        self._step_time = 0.0
+
+    # ------------------------------------------------------------------
+    @property
+    def is_first_stage(self) -> bool:
+        return self.grid.is_first_stage()
+
+    @property
+    def is_last_stage(self) -> bool:
+        return self.grid.is_last_stage()
+
+    def _next_batch(self, data_iter: Iterator):
+        batch = next(data_iter)
+        out = {}
+        for k, v in batch.items():
+            if torch.is_tensor(v):
+                out[k] = v.to(self.device, non_blocking=True)
+            else:
+                out[k] = v
+        return out
+
+    # -- microbatch fwd/bwd ------------------------------------------------
+    def _forward_step(self, recv_act: Optional[torch.Tensor], data_iter):
+        """Returns (input_tensor_for_grad, backward_handle, loss_detached)."""
+        if self.is_first_stage:
+            batch = self._next_batch(data_iter)
+            x = batch["input_ids"]
+            inp = None
+            first_batch = batch
+        else:
+            x = recv_act
+            x.requires_grad_(True)
+            inp = x
+            first_batch = None
+
+        out = self.module(x)
+
+        if self.is_last_stage:
+            if self.is_first_stage:
+                labels = first_batch["labels"]
+            else:
+                labels = self._next_batch(data_iter)["labels"]
+            loss = self.module.loss_fn(out, labels)
+            scaled = loss / self.micro_batches
+            if self.loss_scaler is not None:
+                scaled = scaled * self.loss_scaler.scale
+            return inp, scaled, loss.detach()
+        return inp, out, None
+
+    def _backward_step(self, inp, handle, recv_grad) -> Optional[torch.Tensor]:
+        if self.is_last_stage:
+            handle.backward()
+        else:
+            torch.autograd.backward(handle, grad_tensors=recv_grad)
+        if inp is None:
+            return None
+        g = inp.grad
+        inp.grad = None
+        return g
+
+    # -- the 1F1B schedule --------------------------------------------------
+    def train_batch(self, data_iter: Iterator) -> torch.Tensor:
+        """One optimizer step == ``micro_batches`` microbatches, 1F1B."""
+        t0 = time.time()
+        self.module.train()
+        M = self.micro_batches
+        P = self.grid.num_stages
+        s = self.grid.stage_id
+        warmup = min(P - 1 - s, M)
+        remaining = M - warmup
+
+        pending = deque()  # (input_tensor, backward_handle)
+        losses = []
+
+        # ---- warmup forwards
+        for _ in range(warmup):
+            x = self.p2p.recv_forward()
+            inp, handle, loss = self._forward_step(x, data_iter)
+            self.p2p.send_forward(handle if not self.is_last_stage else None)
+            pending.append((inp, handle))
+
+        x = self.p2p.recv_forward() if remaining > 0 else None
+
+        # ---- steady 1F1B
+        for i in range(remaining):
+            inp, handle, loss = self._forward_step(x, data_iter)
+            pending.append((inp, handle))
+            if loss is not None:
+                losses.append(loss)
+            if self.is_last_stage:
+                recv_grad = None
+            else:
+                recv_grad = self.p2p.send_forward_recv_backward(handle)
+            b_inp, b_handle = pending.popleft()
+            g = self._backward_step(b_inp, b_handle, recv_grad)
+            last = i == remaining - 1
+            if g is None:  # first stage sends nothing backward
+                x = None if last else self.p2p.recv_forward()
+            elif last:
+                self.p2p.send_backward(g)
+            else:
+                x = self.p2p.send_backward_recv_forward(g)
+
+        # ---- cooldown backwards
+        for _ in range(warmup):
+            b_inp, b_handle = pending.popleft()
+            recv_grad = self.p2p.recv_backward() if not self.is_last_stage else None
+            g = self._backward_step(b_inp, b_handle, recv_grad)
+            if g is not None:
+                self.p2p.send_backward(g)
+
+        # ---- boundary: DP all-reduce, clip, step
+        self._optimizer_step()
+
+        loss_out = self._reduce_loss(losses)
+        self._step_time = time.time() - t0
+        return loss_out
+
+    # ------------------------------------------------------------------
+    def _reduce_loss(self, losses) -> torch.Tensor:
+        if self.is_last_stage:
+            loss = torch.stack(losses).mean() if losses else torch.zeros((), device=self.device)
+            loss = loss.float()
+            if self.grid.dp_degree > 1 and dist.is_initialized():
+                # SUM+div (AVG is NCCL-only; gloo path must work for CPU tests)
+                dist.all_reduce(loss, op=dist.ReduceOp.SUM, group=self.grid.dp_group)
+                loss /= self.grid.dp_degree
+        else:
+            loss = torch.zeros((), dtype=torch.float32, device=self.device)
+        if self.grid.num_stages > 1 and dist.is_initialized():
+            src = self.grid.stage_to_rank(self.grid.num_stages - 1)
+            dist.broadcast(loss, src=src, group=self.grid.pipe_group)
+        return loss
+
+    def _allreduce_gradients(self) -> None:
+        if self.grid.dp_degree <= 1 or not dist.is_initialized():
+            return
+        flat = self.optimizer.flat_grads
+        bucket_elems = max(1, self.config.allreduce_bucket_mb * 1024 * 1024 // 4)
+        handles = []
+        for off in range(0, flat.numel(), bucket_elems):
+            chunk = flat.narrow(0, off, min(bucket_elems, flat.numel() - off))
+            handles.append(
+                dist.all_reduce(chunk, op=dist.ReduceOp.SUM, group=self.grid.dp_group,
+                                async_op=True)
+            )
+        for h in handles:
+            h.wait()
+        flat.div_(self.grid.dp_degree)
+
+    def _optimizer_step(self) -> None:
+        self._allreduce_gradients()
+
+        inv_scale = 1.0
+        if self.loss_scaler is not None:
+            inv_scale = 1.0 / self.loss_scaler.scale
+
+        sq = self.optimizer.grad_sq_sum()
+        if self.grid.num_stages > 1 and dist.is_initialized():
+            dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.grid.pipe_group)
+        global_norm = (sq.float().sqrt() * inv_scale).item()
+
+        if self.loss_scaler is not None:
+            found_inf = not (global_norm == global_norm and global_norm != float("inf"))
+            self.loss_scaler.update(found_inf)
+            if found_inf:
+                self.optimizer.zero_grad()
+                self.skipped_steps += 1
+                self.global_steps += 1
+                return
+
+        clip = self.config.optimizer.max_grad_norm
+        coef = inv_scale
+        if clip > 0 and global_norm > clip:
+            coef *= clip / (global_norm + 1e-6)
+        self.optimizer.step(grad_scale=coef)
+        self.optimizer.zero_grad()
+        self.lr_scheduler.step()
+        self.global_steps += 1
+
+    # ------------------------------------------------------------------
+    @property
+    def last_step_time(self) -> float:
+        return self._step_time
+
+    def get_lr(self) -> float:
+        return self.optimizer.lr
+
+    # checkpoint integration lives in lpp_amd.checkpoint
+    def state_dict_local(self) -> dict:
+        return {
+            "optimizer": self.optimizer.state_dict(),
+            "lr_scheduler": self.lr_scheduler.state_dict(),
+            "global_steps": self.global_steps,
+            "skipped_steps": self.skipped_steps,
+            "loss_scale": self.loss_scaler.scale if self.loss_scaler else None,
+        }
+
+    def load_state_dict_local(self, sd: dict) -> None:
+        self.optimizer.load_state_dict(sd["optimizer"])
+        self.lr_scheduler.load_state_dict(sd["lr_scheduler"])
+        self.global_steps = sd.get("global_steps", 0)
+        self.skipped_steps = sd.get("skipped_steps", 0)
+        if self.loss_scaler is not None and sd.get("loss_scale"):
+            self.loss_scaler.scale = sd["loss_scale"]

@@ -1,0 +1,242 @@
+"""LLaMA as a flat list of pipeline-stage layer modules.
+
+Native re-design of the reference's models/llama_ds_mp_wrap.py ("family B":
+``get_layers_from_config`` at :209-224 building LayerSpecs for
+EmbeddingPipe :128, ParallelTransformerLayerPipe :135, LayerNormPipe :184,
+LMLayerPipe :191, with ``loss_fn`` :105-116), with the reference's contract
+quirks fixed:
+
+- The inter-stage tensor is ONE bf16 hidden-state tensor [B, S, H].
+  No [B,1,S,S] additive mask is ever built or shipped (the reference sends
+  the O(S^2) mask through every p2p hop — data/flan.py:194-243,
+  models/llama_ds_mp_wrap.py:37,76,148); causality is implicit in the
+  attention kernel.  No position_ids travel either: each stage regenerates
+  RoPE phases from its static cos/sin table.
+- Labels never smuggle an index column (Q2); the collator emits
+  ``(input_ids, labels)`` and only the last stage consumes labels.
+
+State-dict keys match HF ``LlamaForCausalLM`` per-layer keys with the
+``model.layers.{i}.`` prefix stripped, so convert2ckpt's on-disk layout
+(convert2ckpt.py:19-48) round-trips against real HF weights.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from ..config import ModelConfig
+from ..layer_spec import LayerSpec
+from .. import ops
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden_size: int, eps: float = 1e-6):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.variance_epsilon = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rmsnorm(x, self.weight, self.variance_epsilon)
+
+
+class LlamaAttention(nn.Module):
+    """Self-attention with RoPE; q/k/v/o projections named for HF key parity."""
+
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.num_heads = cfg.num_heads
+        self.num_kv_heads = cfg.kv_heads
+        self.head_dim = cfg.head_dim
+        self.hidden_size = cfg.hidden_size
+        self.rope_theta = cfg.rope_theta
+        self.max_seq_len = cfg.max_seq_len
+        bias = False
+        self.q_proj = nn.Linear(cfg.hidden_size, self.num_heads * self.head_dim, bias=bias)
+        self.k_proj = nn.Linear(cfg.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
+        self.v_proj = nn.Linear(cfg.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
+        self.o_proj = nn.Linear(self.num_heads * self.head_dim, cfg.hidden_size, bias=bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, S, _ = x.shape
+        q = self.q_proj(x).view(B, S, self.num_heads, self.head_dim)
+        k = self.k_proj(x).view(B, S, self.num_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, S, self.num_kv_heads, self.head_dim)
+        cos, sin = ops.build_rope_cache(
+            self.max_seq_len, self.head_dim, self.rope_theta, x.device
+        )
+        q = ops.apply_rope(q, cos, sin)
+        k = ops.apply_rope(k, cos, sin)
+        o = ops.causal_attention(q, k, v)  # [B,S,H,D]
+        return self.o_proj(o.reshape(B, S, self.num_heads * self.head_dim))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class DecoderLayerPipe(nn.Module):
+    """One LLaMA decoder layer; hidden in -> hidden out.
+
+    Mirrors ParallelTransformerLayerPipe (models/llama_ds_mp_wrap.py:135-181)
+    minus the tuple/mask plumbing.  Activation checkpointing recomputes the
+    whole layer (interval semantics handled by the engine).
+    """
+
+    def __init__(self, cfg: ModelConfig, activation_checkpointing: bool = False):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg)
+        self.mlp = LlamaMLP(cfg)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.activation_checkpointing = activation_checkpointing
+
+    def _forward_impl(self, hidden: torch.Tensor) -> torch.Tensor:
+        hidden = hidden + self.self_attn(self.input_layernorm(hidden))
+        hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
+        return hidden
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        if self.activation_checkpointing and self.training and hidden.requires_grad:
+            return torch.utils.checkpoint.checkpoint(
+                self._forward_impl, hidden, use_reentrant=False, preserve_rng_state=False
+            )
+        return self._forward_impl(hidden)
+
+    @staticmethod
+    def spec_param_count(cfg: ModelConfig, activation_checkpointing: bool = False) -> int:
+        h, i = cfg.hidden_size, cfg.intermediate_size
+        kv = cfg.kv_heads * cfg.head_dim
+        return h * h * 2 + 2 * h * kv + 3 * h * i + 2 * h
+
+
+class EmbeddingPipe(nn.Embedding):
+    """Stage-0 entry: input_ids [B,S] int64 -> hidden [B,S,H].
+    (reference: EmbeddingPipe, models/llama_ds_mp_wrap.py:128-132)"""
+
+    def __init__(self, vocab_size: int, hidden_size: int):
+        super().__init__(vocab_size, hidden_size)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        return super().forward(input_ids)
+
+    @staticmethod
+    def spec_param_count(vocab_size: int, hidden_size: int) -> int:
+        return vocab_size * hidden_size
+
+
+class NormPipe(RMSNorm):
+    """Final RMSNorm (reference: LayerNormPipe, models/llama_ds_mp_wrap.py:184-188)."""
+
+    @staticmethod
+    def spec_param_count(hidden_size: int, eps: float = 1e-6) -> int:
+        return hidden_size
+
+
+class LMHeadPipe(nn.Linear):
+    """LM head (reference: LMLayerPipe, models/llama_ds_mp_wrap.py:191-195).
+    Untied from the embedding by design (README.md:44-46)."""
+
+    def __init__(self, hidden_size: int, vocab_size: int):
+        super().__init__(hidden_size, vocab_size, bias=False)
+
+    @staticmethod
+    def spec_param_count(hidden_size: int, vocab_size: int) -> int:
+        return hidden_size * vocab_size
+
+
+def loss_fn(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Shift-by-one causal CE (reference loss_fn, models/llama_ds_mp_wrap.py:105-116).
+    Runs only on the last stage."""
+    return ops.shifted_cross_entropy(logits, labels)
+
+
+def get_layers_from_config(
+    cfg: ModelConfig, activation_checkpointing: bool = False
+) -> List[LayerSpec]:
+    """Flat LayerSpec list: embedding, L decoder layers, final norm, LM head.
+    Index i of this list == layer file number in the checkpoint layout
+    (convert2ckpt.py:23-36)."""
+    specs: List[LayerSpec] = [LayerSpec(EmbeddingPipe, cfg.vocab_size, cfg.hidden_size)]
+    for _ in range(cfg.num_layers):
+        specs.append(LayerSpec(DecoderLayerPipe, cfg, activation_checkpointing))
+    specs.append(LayerSpec(NormPipe, cfg.hidden_size, cfg.rms_norm_eps))
+    specs.append(LayerSpec(LMHeadPipe, cfg.hidden_size, cfg.vocab_size))
+    return specs
+
+
+@torch.no_grad()
+def init_weights(module: nn.Module, cfg: ModelConfig, generator: Optional[torch.Generator] = None):
+    """HF-style init: normal(0, initializer_range) for weight matrices,
+    ones for norms.  Deterministic given the generator."""
+    std = cfg.initializer_range
+    for m in module.modules():
+        if isinstance(m, nn.Linear):
+            m.weight.normal_(0.0, std, generator=generator)
+            if m.bias is not None:
+                m.bias.zero_()
+        elif isinstance(m, nn.Embedding):
+            m.weight.normal_(0.0, std, generator=generator)
+        elif isinstance(m, RMSNorm):
+            m.weight.fill_(1.0)
+
+
+@torch.no_grad()
+def deterministic_layer_init(layer: nn.Module, cfg: ModelConfig, seed: int, global_idx: int):
+    """Seed per GLOBAL layer index so a pipeline stage initialises exactly the
+    weights the monolithic model has for that layer — regardless of how the
+    stage boundaries fall.  This is the oracle-alignment mechanism for the
+    PP-vs-monolithic loss-equivalence tests."""
+    g = torch.Generator().manual_seed(seed * 100003 + global_idx)
+    std = cfg.initializer_range
+    for name, p in sorted(layer.named_parameters()):
+        if p.dim() >= 2:
+            p.copy_(torch.empty(p.shape, dtype=torch.float32).normal_(0.0, std, generator=g).to(p.dtype))
+        else:
+            p.fill_(1.0) if "norm" in name or isinstance(layer, (RMSNorm,)) else p.zero_()
+    # norm weights inside decoder layers
+    for m in layer.modules():
+        if isinstance(m, RMSNorm):
+            m.weight.fill_(1.0)
+
+
+def init_pipeline_weights(pipeline_module, cfg: ModelConfig, seed: int) -> None:
+    for local_idx, layer in enumerate(pipeline_module.layers):
+        deterministic_layer_init(layer, cfg, seed, pipeline_module.global_layer_index(local_idx))
+
+
+def init_monolithic_weights(model: "LlamaForCausalLM", seed: int) -> None:
+    for gidx, layer in enumerate(model.layers):
+        deterministic_layer_init(layer, model.cfg, seed, gidx)
+
+
+class LlamaForCausalLM(nn.Module):
+    """Monolithic (non-pipeline) model built from the same layer specs —
+    the numerics oracle for PP-vs-single-process loss-equivalence tests
+    (SURVEY.md §4)."""
+
+    def __init__(self, cfg: ModelConfig, activation_checkpointing: bool = False):
+        super().__init__()
+        self.cfg = cfg
+        self.layers = nn.ModuleList(
+            [s.build() for s in get_layers_from_config(cfg, activation_checkpointing)]
+        )
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        x = input_ids
+        for layer in self.layers:
+            x = layer(x)
+        return x
+
+    def compute_loss(self, input_ids: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+        return loss_fn(self.forward(input_ids), labels)

@@ -1,0 +1,117 @@
+"""Process grid: rank <-> (pipeline stage, data-parallel id).
+
+Native replacement for DeepSpeed's PipeDataParallelTopology / PipelineModule
+grid, whose queries the reference training loop depends on:
+``model.grid.get_data_parallel_id()`` (trainer_base_ds_mp.py:313) and
+``is_first_stage() / is_last_stage()`` (trainer_base_ds_mp.py:309).
+
+Layout choice (MI355X single node, fully-connected xGMI mesh): stage-major —
+``rank = stage * dp_degree + dp_id``.  Neighbouring pipeline stages for a
+given dp column are ``rank ± dp_degree``; every pair of GPUs on a node has a
+dedicated ~153 GB/s xGMI link, so any placement is one hop — stage-major keeps
+the DP all-reduce group ({stage*dp .. stage*dp+dp-1}) contiguous, which is
+what RCCL's communicator setup prefers.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch.distributed as dist
+
+
+@dataclass
+class _Groups:
+    pipe: object = None  # this rank's pipeline group (its dp column)
+    data: object = None  # this rank's DP group (its stage row)
+
+
+class ProcessGrid:
+    """world_size = num_stages * dp_degree, stage-major rank layout."""
+
+    def __init__(self, world_size: int, rank: int, num_stages: int):
+        if num_stages <= 0:
+            raise ValueError("num_stages must be positive")
+        if world_size % num_stages != 0:
+            raise ValueError(
+                f"world_size {world_size} not divisible by num_stages {num_stages} "
+                "(hybrid PP x DP requires world_size = num_stages * dp_degree; "
+                "reference math at trainer_base_ds_mp.py:245)"
+            )
+        self.world_size = world_size
+        self.rank = rank
+        self.num_stages = num_stages
+        self.dp_degree = world_size // num_stages
+        self.stage_id = rank // self.dp_degree
+        self.dp_id = rank % self.dp_degree
+        self._groups = _Groups()
+
+    # -- queries used by the training loop (reference parity) -------------
+    def is_first_stage(self) -> bool:
+        return self.stage_id == 0
+
+    def is_last_stage(self) -> bool:
+        return self.stage_id == self.num_stages - 1
+
+    def get_data_parallel_id(self) -> int:
+        return self.dp_id
+
+    def get_pipe_parallel_rank(self) -> int:
+        return self.stage_id
+
+    def get_data_parallel_world_size(self) -> int:
+        return self.dp_degree
+
+    def get_pipe_parallel_world_size(self) -> int:
+        return self.num_stages
+
+    # -- neighbours --------------------------------------------------------
+    def stage_to_rank(self, stage: int, dp_id: Optional[int] = None) -> int:
+        dp = self.dp_id if dp_id is None else dp_id
+        return stage * self.dp_degree + dp
+
+    @property
+    def prev_rank(self) -> Optional[int]:
+        if self.is_first_stage():
+            return None
+        return self.stage_to_rank(self.stage_id - 1)
+
+    @property
+    def next_rank(self) -> Optional[int]:
+        if self.is_last_stage():
+            return None
+        return self.stage_to_rank(self.stage_id + 1)
+
+    # -- process groups ----------------------------------------------------
+    def build_groups(self) -> None:
+        """Create the pipe/data subgroups.  Must be called on ALL ranks with
+        identical arguments (dist.new_group is collective)."""
+        if not dist.is_initialized():
+            return
+        # DP groups: one per stage row.
+        for stage in range(self.num_stages):
+            ranks = [stage * self.dp_degree + d for d in range(self.dp_degree)]
+            g = dist.new_group(ranks=ranks)
+            if self.rank in ranks:
+                self._groups.data = g
+        # Pipe groups: one per dp column.
+        for dp in range(self.dp_degree):
+            ranks = [s * self.dp_degree + dp for s in range(self.num_stages)]
+            g = dist.new_group(ranks=ranks)
+            if self.rank in ranks:
+                self._groups.pipe = g
+
+    @property
+    def dp_group(self):
+        return self._groups.data
+
+    @property
+    def pipe_group(self):
+        return self._groups.pipe
+
+    def __repr__(self) -> str:
+        return (
+            f"ProcessGrid(rank={self.rank}/{self.world_size}, "
+            f"stage={self.stage_id}/{self.num_stages}, dp={self.dp_id}/{self.dp_degree})"
+        )

@@ -1,0 +1,50 @@
+"""Multi-process pipeline tests on gloo (BASELINE config #1: plumbing on CPU,
+world_size=2 — SURVEY.md §4 'Schedule/deadlock tests without GPUs')."""
+
+import pytest
+import torch
+
+from tests.dist_utils import run_dist
+from tests.engine_utils import run_steps
+
+
+def _single_process_baseline(steps=3, gas=4):
+    return run_steps(rank=0, world_size=1, num_stages=1, steps=steps, gas=gas)
+
+
+def test_pp2_matches_single_process():
+    """PP=2 over gloo must reproduce the monolithic loss trajectory exactly
+    (fp32, identical per-layer weights + data order)."""
+    base = _single_process_baseline()
+    got = run_dist(2, run_steps, 2, 3, 4)  # num_stages=2, steps=3, gas=4
+    for r in range(2):
+        for a, b in zip(base, got[r]):
+            assert abs(a - b) < 1e-4, (base, got[r])
+
+
+def test_pp2_more_microbatches_than_stages():
+    """gas > stages exercises the steady-state 1F1B path."""
+    base = _single_process_baseline(steps=2, gas=6)
+    got = run_dist(2, run_steps, 2, 2, 6)
+    for a, b in zip(base, got[0]):
+        assert abs(a - b) < 1e-4
+
+
+def test_pp2_fewer_microbatches_than_depth():
+    """gas=1: pure fill-drain, no steady state (warmup==M edge case)."""
+    base = _single_process_baseline(steps=2, gas=1)
+    got = run_dist(2, run_steps, 2, 2, 1)
+    for a, b in zip(base, got[0]):
+        assert abs(a - b) < 1e-4
+
+
+def test_dp2_pure_data_parallel():
+    """PP=1 x DP=2: both ranks return the identical (dp-averaged) loss and
+    stay in sync across steps."""
+    got = run_dist(2, run_steps, 1, 3, 4)
+    assert got[0] == pytest.approx(got[1], abs=1e-6)
+
+
+def test_pp2_all_ranks_same_loss():
+    got = run_dist(2, run_steps, 2, 2, 4)
+    assert got[0] == pytest.approx(got[1], abs=1e-6)
