@@ -1,0 +1,195 @@
+#!/usr/bin/env python
+"""Flagship benchmark: LLaMA-65B pipeline-parallel training throughput on
+MI355X (BASELINE.json metric: "tokens/sec (whole node), LLaMA-65B PP=8
+seq4096 at 1/2/4/8 MI355X").
+
+Weak scaling over pipeline depth: each GPU holds one 65B-shaped stage
+(hidden 8192, heads 64, intermediate 22016 — 10 of the 80 decoder layers
+per stage).  At --gpus 8 this is the full LLaMA-65B with PP=8; at smaller N
+the model is the first N/8 slice of it, so per-GPU work is fixed as N grows
+and the ideal scaling curve is FLAT tokens/s (the model grows with N; any
+drop measures pipeline bubble + p2p cost).
+
+Synthetic data (random tokens of the BASELINE shape), random-init weights,
+bf16 compute, fp32 grad accumulation + master AdamW, activation
+checkpointing interval 1 — nothing is skipped inside the timed region: each
+step runs `gas` microbatches forward+backward plus the full optimizer step.
+
+Launch (the driver's contract):
+  python bench.py --gpus 1 --steps 3 --warmup 1
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+from lpp_amd.config import TrainConfig, model_config
+from lpp_amd.data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset
+from lpp_amd.engine import PipelineEngine
+from lpp_amd.models import DecoderLayerPipe, EmbeddingPipe, RMSNorm, get_layers_from_config, loss_fn
+from lpp_amd.pipeline_module import PipelineModule
+from lpp_amd.topology import ProcessGrid
+from lpp_amd.utils import init_distributed, set_seed
+
+
+@torch.no_grad()
+def fast_random_init(module: PipelineModule, std: float, seed: int) -> None:
+    """On-device random init (normal(0, std) matrices, ones for norms).
+    Per-param generator seeding keeps it deterministic per rank."""
+    dev = next(module.parameters()).device
+    g = torch.Generator(device=dev)
+    g.manual_seed(seed + 1000 * module.grid.rank)
+    for m in module.modules():
+        if isinstance(m, RMSNorm):
+            m.weight.fill_(1.0)
+        elif isinstance(m, (torch.nn.Linear, torch.nn.Embedding)):
+            m.weight.normal_(0.0, std, generator=g)
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--gas", type=int, default=32, help="microbatches per step")
+    ap.add_argument("--micro-batch-size", type=int, default=1)
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--model", type=str, default="llama-65b")
+    ap.add_argument("--layers-per-stage", type=int, default=10,
+                    help="decoder layers per GPU (65B/8 = 10)")
+    ap.add_argument("--dp", type=int, default=1, help="data-parallel degree")
+    ap.add_argument("--dtype", type=str, default="bf16")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    n_gpus = max(world, 1)
+    if world > 1:
+        init_distributed()
+
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    num_stages = world // args.dp if world > 1 else 1
+    dp = args.dp if world > 1 else 1
+    # weak scaling: model depth grows with pipeline depth
+    num_layers = args.layers_per_stage * num_stages
+    mcfg = model_config(args.model, num_layers=num_layers, max_seq_len=args.seq_len)
+
+    cfg = TrainConfig(
+        model=mcfg,
+        num_stages=num_stages,
+        micro_batch_size=args.micro_batch_size,
+        gradient_accumulation_steps=args.gas,
+        seq_len=args.seq_len,
+        dtype=args.dtype if on_gpu else "fp32",
+        activation_checkpoint_interval=1,
+    )
+    cfg.optimizer.lr = 1e-5
+    cfg.optimizer.total_num_steps = 1000
+
+    set_seed(1234, rank)
+    grid = ProcessGrid(max(world, 1), rank, num_stages)
+    grid.build_groups()
+
+    from lpp_amd.config import torch_dtype
+
+    module = PipelineModule(
+        get_layers_from_config(mcfg),
+        grid,
+        loss_fn=loss_fn,
+        activation_checkpoint_interval=cfg.activation_checkpoint_interval,
+        device=device,
+        dtype=torch_dtype(cfg.dtype),
+    )
+    fast_random_init(module, mcfg.initializer_range, seed=1234)
+    engine = PipelineEngine(module, cfg, grid, device=device)
+
+    # data: only first/last stage iterate
+    n_examples = args.gas * args.micro_batch_size * (args.steps + args.warmup + 1)
+    ds = SyntheticCausalLMDataset(n_examples, args.seq_len, mcfg.vocab_size, seed=7)
+    loader = torch.utils.data.DataLoader(
+        ds, batch_size=args.micro_batch_size, shuffle=False,
+        collate_fn=CausalLMCollator(args.seq_len), drop_last=True,
+    )
+    it = iter(RepeatingLoader(loader))
+
+    def sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine.train_batch(it)
+
+    sync()
+    t0 = time.time()
+    last_loss = 0.0
+    for _ in range(args.steps):
+        last_loss = float(engine.train_batch(it))
+    sync()
+    elapsed = time.time() - t0
+
+    # MAX elapsed over ranks (ranks are barrier-synced; take max anyway)
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized():
+        t_dev = t.to(device) if on_gpu else t
+        dist.all_reduce(t_dev, op=dist.ReduceOp.MAX)
+        elapsed = float(t_dev.item())
+
+    tokens_per_step = args.gas * args.micro_batch_size * args.seq_len * dp
+    tokens_per_sec = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        par = f"pp{num_stages}" + (f"_dp{dp}" if dp > 1 else "")
+        result = {
+            "metric": "tokens/sec (whole node), LLaMA-65B PP=8 seq4096 at 1/2/4/8 MI355X",
+            "value": round(tokens_per_sec, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": cfg.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.gas * args.micro_batch_size * dp,
+                "seq_len": args.seq_len,
+                "parallelism": par,
+                "layers_per_stage": args.layers_per_stage,
+                "num_layers": num_layers,
+                "micro_batch_size": args.micro_batch_size,
+                "grad_accum_steps": args.gas,
+                "activation_checkpointing": True,
+                "last_loss": round(last_loss, 4),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

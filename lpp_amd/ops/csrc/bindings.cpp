@@ -1,0 +1,33 @@
+// Python bindings for the lpp_amd gfx950 kernel extension.
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor weight, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
+                                    at::Tensor invrms);
+at::Tensor rope_fwd(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, int64_t pos_offset);
+at::Tensor rope_bwd(at::Tensor dy, at::Tensor cos_t, at::Tensor sin_t, int64_t pos_offset);
+at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up);
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor gate, at::Tensor up);
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor labels);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
+                             double scale);
+void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> masters,
+                 std::vector<at::Tensor> grads, std::vector<at::Tensor> exp_avg,
+                 std::vector<at::Tensor> exp_avg_sq, double lr, double beta1, double beta2,
+                 double eps, double weight_decay, double bias1, double bias2,
+                 double grad_scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "lpp_amd gfx950 (MI355X/CDNA4) kernels";
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, invrms)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (dx, dw)");
+  m.def("rope_fwd", &rope_fwd, "RoPE forward");
+  m.def("rope_bwd", &rope_bwd, "RoPE backward");
+  m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (dgate, dup)");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward (loss_sum, lse, count)");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward (dlogits)");
+  m.def("fused_adamw", &fused_adamw, "fused mixed-precision AdamW");
+}

@@ -47,9 +47,19 @@ class PipelineModule(nn.Module):
         self.local_stop = self.bounds[s + 1]
 
         built: List[nn.Module] = []
-        for spec in self.specs[self.local_start : self.local_stop]:
-            m = spec.build() if isinstance(spec, LayerSpec) else spec
-            built.append(m)
+        prev_dtype = torch.get_default_dtype()
+        try:
+            if dtype is not None:
+                torch.set_default_dtype(dtype)
+            import contextlib
+
+            ctx = torch.device(device) if device is not None else contextlib.nullcontext()
+            with ctx:
+                for spec in self.specs[self.local_start : self.local_stop]:
+                    m = spec.build() if isinstance(spec, LayerSpec) else spec
+                    built.append(m)
+        finally:
+            torch.set_default_dtype(prev_dtype)
         self.layers = nn.ModuleList(built)
         if device is not None or dtype is not None:
             self.layers.to(device=device, dtype=dtype)
