@@ -42,6 +42,16 @@ def layer_file(idx: int) -> str:
     return LAYER_FMT.format(idx=idx)
 
 
+def normalize_ckpt_path(ckpt_dir: str, tag: Optional[str]) -> tuple[str, Optional[str]]:
+    """Accept either the checkpoint root (with a ``latest`` tag file) or a
+    step dir directly (``.../global_step4`` — how the reference's cfg.resume
+    names it, trainer_base_ds_mp.py:452-455)."""
+    p = Path(ckpt_dir)
+    if tag is None and not (p / LATEST).exists() and any(p.glob("layer_*-model_*.pt")):
+        return str(p.parent), p.name
+    return ckpt_dir, tag
+
+
 def read_latest(ckpt_dir: str) -> Optional[str]:
     p = Path(ckpt_dir) / LATEST
     if not p.exists():
@@ -100,6 +110,7 @@ def load_module_weights(module, ckpt_dir: str, tag: Optional[str] = None,
                         strict: bool = True, dtype: Optional[torch.dtype] = None) -> str:
     """Module-only warm start: map this stage's layer files onto local layers.
     Works on converted-HF dirs (no optimizer state present)."""
+    ckpt_dir, tag = normalize_ckpt_path(ckpt_dir, tag)
     tag = tag or read_latest(ckpt_dir)
     if tag is None:
         raise FileNotFoundError(f"no 'latest' tag in {ckpt_dir}")
@@ -125,6 +136,7 @@ def load_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
     (load_module_only=True — the reference's load path at
     trainer_base_ds_mp.py:284 with the same flag)."""
     grid = engine.grid
+    ckpt_dir, tag = normalize_ckpt_path(ckpt_dir, tag)
     tag = tag or read_latest(ckpt_dir)
     if tag is None:
         raise FileNotFoundError(f"no 'latest' tag in {ckpt_dir}")

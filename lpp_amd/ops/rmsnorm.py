@@ -14,10 +14,12 @@ from . import use_hip, extension
 
 
 def rmsnorm_ref(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
-    """Eager reference: fp32 math, cast back to input dtype."""
+    """Eager reference: full fp32 math, ONE rounding to the input dtype at
+    the end (matches the HIP kernel; HF's LlamaRMSNorm double-rounds by
+    casting before the weight multiply)."""
     xf = x.float()
     inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
-    return (xf * inv).to(x.dtype) * weight
+    return (xf * inv * weight.float()).to(x.dtype)
 
 
 class _RMSNormHIP(torch.autograd.Function):

@@ -1,0 +1,200 @@
+"""Training loop + CLI — native equivalent of trainer_base_ds_mp.py.
+
+Capabilities mirrored from the reference (file:line cites into
+/root/reference/trainer_base_ds_mp.py):
+- distributed init + device pinning (:397-399)
+- engine + module build from config, per-stage weight warm start (:280-299)
+- per-(stage, dp) dataloaders; only first/last stage load data (:309-336)
+- step-count agreement across ranks — computed once from config instead of
+  per-rank len(dataloader)//gas (quirk Q3)
+- resume with dataloader fast-forward (:345-351)
+- periodic checkpoint save + latest tag (:203-224, 367-371)
+- rank-0 logging of {loss, lr}; wandb if available (:360-374,441-447)
+
+CLI:  python -m lpp_amd.trainer --config conf/llama_65b_pp8.yaml [k=v ...]
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import sys
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .checkpoint import load_engine_checkpoint, parse_checkpoint_step, save_engine_checkpoint
+from .config import TrainConfig, torch_dtype
+from .data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset, build_loader
+from .engine import PipelineEngine
+from .models import get_layers_from_config, init_pipeline_weights, loss_fn
+from .pipeline_module import PipelineModule
+from .topology import ProcessGrid
+from .utils import init_distributed, set_seed
+
+logger = logging.getLogger("lpp_amd.trainer")
+
+
+def build_dataset(cfg: TrainConfig):
+    """Synthetic dataset by default (no-network environment); a real corpus
+    drops in behind the same dict contract."""
+    n = cfg.total_dataset_len or 4096
+    return SyntheticCausalLMDataset(n, cfg.seq_len, cfg.model.vocab_size, seed=cfg.seed)
+
+
+def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dict:
+    grid = engine.grid
+    rank0 = grid.rank == 0
+
+    wandb = None
+    if rank0 and os.environ.get("WANDB_MODE", "") not in ("", "disabled", "offline-off"):
+        try:
+            import wandb as _wandb
+
+            _wandb.init(project=os.environ.get("WANDB_PROJECT", "lpp-amd"),
+                        config=cfg.to_dict())
+            wandb = _wandb
+        except Exception:
+            wandb = None
+
+    needs_data = grid.is_first_stage() or grid.is_last_stage()
+    dataset = build_dataset(cfg) if needs_data else None
+
+    # quirk Q3 fix: every rank derives the identical step count from config,
+    # not from its local dataloader length.
+    n_examples = cfg.total_dataset_len or (len(dataset) if dataset is not None else 4096)
+    examples_per_step = cfg.micro_batch_size * cfg.gradient_accumulation_steps * grid.dp_degree
+    steps_per_epoch = n_examples // examples_per_step
+    total_steps = cfg.max_steps or steps_per_epoch * cfg.num_train_epochs
+    cfg.optimizer.total_num_steps = max(cfg.optimizer.total_num_steps, total_steps)
+    engine.lr_scheduler.total_num_steps = cfg.optimizer.total_num_steps
+
+    if rank0:
+        logger.info(
+            "training: %d steps (%d/epoch), global batch %d, grid %s",
+            total_steps, steps_per_epoch, examples_per_step, grid,
+        )
+        os.makedirs(cfg.output_dir, exist_ok=True)
+        cfg.save(os.path.join(cfg.output_dir, "training_config.yaml"))
+
+    step = 0
+    tr_loss = 0.0
+    t_start = time.time()
+    done = False
+    for epoch in range(cfg.num_train_epochs):
+        if done:
+            break
+        it = None
+        if needs_data:
+            loader = build_loader(
+                dataset, cfg.micro_batch_size, grid.dp_degree, grid.dp_id,
+                seed=cfg.seed, num_workers=cfg.num_workers,
+                collator=CausalLMCollator(cfg.seq_len), epoch=epoch,
+            )
+            it = iter(RepeatingLoader(loader))
+        for _ in range(steps_per_epoch):
+            if step >= total_steps:
+                done = True
+                break
+            if step < resume_step:
+                # fast-forward: drain the sampler without compute
+                # (reference :345-351)
+                if needs_data:
+                    for _ in range(cfg.gradient_accumulation_steps):
+                        next(it)
+                step += 1
+                continue
+            loss = engine.train_batch(it)
+            step += 1
+            tr_loss += float(loss)
+            if rank0 and cfg.logging_steps and step % cfg.logging_steps == 0:
+                avg = tr_loss / cfg.logging_steps
+                tr_loss = 0.0
+                msg = {
+                    "step": step,
+                    "loss": round(avg, 4),
+                    "lr": engine.get_lr(),
+                    "s/step": round(engine.last_step_time, 3),
+                }
+                logger.info("%s", msg)
+                if wandb:
+                    wandb.log(msg, step=step)
+            if cfg.save_steps and step % cfg.save_steps == 0:
+                save_engine_checkpoint(
+                    engine, cfg.output_dir, tag=f"global_step{step}",
+                    client_state={"step": step},
+                )
+    final = {"steps": step, "elapsed": time.time() - t_start}
+    if cfg.save_steps:
+        save_engine_checkpoint(engine, cfg.output_dir, tag=f"global_step{step}",
+                               client_state={"step": step})
+    if wandb:
+        wandb.finish()
+    return final
+
+
+def main(argv: Optional[list] = None) -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", type=str, required=True)
+    ap.add_argument("overrides", nargs="*", help="key=value config overrides")
+    args = ap.parse_args(argv)
+
+    cfg = TrainConfig.load(args.config)
+    for ov in args.overrides:
+        k, _, v = ov.partition("=")
+        obj = cfg
+        parts = k.split(".")
+        for p in parts[:-1]:
+            obj = getattr(obj, p)
+        cur = getattr(obj, parts[-1])
+        setattr(obj, parts[-1], type(cur)(v) if cur is not None else v)
+
+    logging.basicConfig(
+        level=logging.INFO,
+        format="%(asctime)s %(name)s [%(levelname)s] %(message)s",
+    )
+
+    rank, world = init_distributed(backend=None if torch.cuda.is_available() else "gloo")
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    set_seed(cfg.seed, rank)
+    grid = ProcessGrid(world, rank, cfg.num_stages)
+    grid.build_groups()
+
+    module = PipelineModule(
+        get_layers_from_config(cfg.model),
+        grid,
+        loss_fn=loss_fn,
+        activation_checkpoint_interval=(
+            cfg.activation_checkpoint_interval if cfg.activation_checkpointing else 0
+        ),
+        device=device,
+        dtype=torch_dtype(cfg.dtype),
+    )
+    init_pipeline_weights(module, cfg.model, seed=cfg.seed)
+    engine = PipelineEngine(module, cfg, grid, device=device)
+
+    resume_step = 0
+    if cfg.model_name_or_path and os.path.isdir(cfg.model_name_or_path):
+        # module-only warm start from a converted HF checkpoint (:284)
+        load_engine_checkpoint(engine, cfg.model_name_or_path, load_module_only=True)
+    if cfg.resume:
+        load_engine_checkpoint(engine, cfg.resume)
+        resume_step = parse_checkpoint_step(cfg.resume) or engine.global_steps
+
+    train(cfg, engine, resume_step=resume_step)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
