@@ -182,6 +182,9 @@ def main() -> int:
                 "grad_accum_steps": args.gas,
                 "activation_checkpointing": True,
                 "last_loss": round(last_loss, 4),
+                "peak_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2)
+                if on_gpu
+                else None,
             },
         }
         print(json.dumps(result), flush=True)
