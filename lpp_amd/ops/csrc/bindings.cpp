@@ -13,6 +13,8 @@ std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor gate, at::Tensor up
 std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor labels);
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
                              double scale);
+std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v);
+at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> masters,
                  std::vector<at::Tensor> grads, std::vector<at::Tensor> exp_avg,
                  std::vector<at::Tensor> exp_avg_sq, double lr, double beta1, double beta2,
@@ -30,4 +32,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward (loss_sum, lse, count)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward (dlogits)");
   m.def("fused_adamw", &fused_adamw, "fused mixed-precision AdamW");
+  m.def("attention_fwd", &attention_fwd, "flash causal attention forward (O, LSE2)");
+  m.def("mfma_test_16x16x32", &mfma_test_16x16x32, "MFMA layout validation");
 }

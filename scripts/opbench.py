@@ -1,4 +1,6 @@
 #!/usr/bin/env python
+import sys, pathlib
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
 """Per-op A/B microbench: HIP kernel vs eager PyTorch on MI355X.
 
 Shapes are the 65B hot-path shapes (hidden 8192, intermediate 22016,
