@@ -35,14 +35,14 @@ def causal_attention_ref(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> t
     return out.transpose(1, 2).contiguous()
 
 
-def _hip_has_attention() -> bool:
+def _hip_supported(q: torch.Tensor) -> bool:
+    """HIP flash kernels cover the production shape: bf16, head_dim 128."""
     if force_eager():
         return False
-    try:
-        ext = extension()
-    except RuntimeError:
+    if q.dtype is not torch.bfloat16 or q.shape[-1] != 128:
         return False
-    return hasattr(ext, "attention_fwd")
+    ext = extension()  # raises loudly if the native path is missing on GPU
+    return hasattr(ext, "attention_fwd") and hasattr(ext, "attention_bwd")
 
 
 class _FlashAttnHIP(torch.autograd.Function):
@@ -62,6 +62,6 @@ class _FlashAttnHIP(torch.autograd.Function):
 
 
 def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
-    if use_hip(q) and _hip_has_attention():
+    if use_hip(q) and _hip_supported(q):
         return _FlashAttnHIP.apply(q.contiguous(), k.contiguous(), v.contiguous())
     return causal_attention_ref(q, k, v)

@@ -14,6 +14,8 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor labels);
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
                              double scale);
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v);
+std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
+                                      at::Tensor v, at::Tensor o, at::Tensor lse2);
 at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> masters,
                  std::vector<at::Tensor> grads, std::vector<at::Tensor> exp_avg,
@@ -33,5 +35,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward (dlogits)");
   m.def("fused_adamw", &fused_adamw, "fused mixed-precision AdamW");
   m.def("attention_fwd", &attention_fwd, "flash causal attention forward (O, LSE2)");
+  m.def("attention_bwd", &attention_bwd, "flash causal attention backward (dQ, dK, dV)");
   m.def("mfma_test_16x16x32", &mfma_test_16x16x32, "MFMA layout validation");
 }
