@@ -32,6 +32,18 @@ def test_mfma_layout(ext):
     assert torch.allclose(C, ref, atol=1e-2, rtol=1e-2), (C - ref).abs().max()
 
 
+def test_mfma_32x32x16_layout(ext):
+    """Validate the 32x32x16 A/B/C fragment layouts (8-wave fwd kernel)."""
+    torch.manual_seed(10)
+    A = torch.randn(32, 16, device=_dev(), dtype=torch.bfloat16)
+    B = (torch.arange(16 * 32, device=_dev(), dtype=torch.float32).view(16, 32) % 7 - 3).to(
+        torch.bfloat16
+    )
+    C = ext.mfma_test_32x32x16(A, B)
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=1e-2, rtol=1e-2), (C - ref).abs().max()
+
+
 def _ref_attn(q, k, v):
     B, S, H, D = q.shape
     Hkv = k.shape[2]
