@@ -6,47 +6,115 @@
 // P = exp2(c*QK^T - LSE2) tile-by-tile from the forward's base-2 logsumexp,
 // so nothing S^2-shaped is ever stored.
 //
-// Standard FA2 decomposition, three kernels:
-//   1. delta[b,h,s]  = rowsum(dO * O)                  (memory-bound)
-//   2. dK/dV kernel  — parallel over KV tiles, loops q tiles >= diag:
-//        P^T  = exp2(c*K Q^T - L[q])
-//        dV  += P^T dO
-//        dP^T = V dO^T
-//        dS^T = P^T * (dP^T - delta[q]);   dK += scale * dS^T Q
-//   3. dQ kernel     — parallel over Q tiles, loops kv tiles <= diag:
-//        P = exp2(c*Q K^T - L);  dP = dO V^T
-//        dS = P * (dP - delta);  dQ += scale * dS K
+// Same swapped-operand MFMA structure as the forward (v_mfma_f32_32x32x16):
+// the softmax-corrected tensors (P, dS) are always computed with the
+// REDUCTION-FREE axis in the lane dimension, so the elementwise
+// dS = P*(dP - delta) needs no cross-lane traffic, and the packed A-
+// fragments for the accumulating GEMMs come from the same
+// v_cvt_pk_bf16_f32 + permlane32_swap construction the forward uses.
 //
-// MFMA 16x16x32 bf16 fragment layouts identical to the forward
-// (validated by mfma_test.hip):
-//   A[m][k]: lane m = lane&15, k = (lane>>4)*8 + j
-//   B[k][n]: lane n = lane&15, k = (lane>>4)*8 + j
-//   C[m][n]: reg r -> m = (lane>>4)*4 + r, n = lane&15
-// LDS staging: "normal" [rows][128] tiles are XOR-swizzled (kswz) for the
-// 16-distinct-row B-frag read; "transposed" [128][rows+8] tiles give the
-// contiguous-k B-frag read (same trick as the forward's K/Vt tiles).
+// Three kernels:
+//   1. delta[b,h,s] = rowsum(dO * O)                       (memory-bound)
+//   2. dQ kernel — 8 waves x 32 q rows (lane = q): per KV tile of 64,
+//        S^T  = mfma(K_lds, Q_reg)   C[kv regs][q lane]
+//        dP^T = mfma(V_lds, dO_reg)  same layout
+//        P    = exp2(c*S^T - L[own q]);  dS = P*(dP^T - delta[own q])
+//        dQ  += mfma(pack(dS), K^T image)          (scale at epilogue)
+//   3. dK/dV kernel — 4 waves x 32 kv rows (lane = kv), 1 wave/SIMD
+//      (512-VGPR budget), grid over KV blocks x B*HKV; per 32-row q tile:
+//        S    = mfma(Q_lds, K_reg)   C[q regs][kv lane]
+//        dP   = mfma(dO_lds, V_reg)  same layout
+//        P    = exp2(c*S - L[q])  (L,delta broadcast from a staged tile)
+//        dV  += mfma(pack(P),  dO^T image)
+//        dK  += mfma(pack(dS), Q^T image)
+//      GQA: the G query heads sharing a kv head accumulate in-register.
+//
+// LDS images: "normal" [rows][128] tiles XOR-swizzled ((row&15)<<4) for the
+// A-fragment ds_read_b128; "transposed" [128][rows+pad] tiles (pad -> odd
+// word stride -> conflict-free B-fragment reads), filled by the same
+// in-register 4x4 dword butterfly as the forward's V image.
 #include "common.h"
 
 namespace lpp {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
-typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(2))) int int2v;
 
 constexpr int AB_D = 128;
 
+__device__ __forceinline__ int bswz(int row, int col_elem) {
+  return row * 256 + ((col_elem * 2) ^ ((row & 15) << 4));
+}
+__device__ __forceinline__ unsigned cvt_pk_bf16_(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+__device__ __forceinline__ float bf2f_(short s) {
+  union { unsigned u; float f; } cv;
+  cv.u = ((unsigned)(unsigned short)s) << 16;
+  return cv.f;
+}
 __device__ __forceinline__ short f2bf_(float f) {
   __hip_bfloat16 h = __float2bfloat16(f);
   return *reinterpret_cast<short*>(&h);
 }
 
-// swizzled byte offset inside a [rows][128] bf16 LDS tile (row stride 256 B)
-__device__ __forceinline__ int bswz(int row, int col_elem) {
-  return row * 256 + ((col_elem * 2) ^ ((row & 15) << 4));
+// 4x4 dword butterfly across the lane quad {l, l^16, l^32, l^48}: lane with
+// quad position p ends holding dword p of each quad row (see the forward).
+// Returns the four dwords (rows 0..3 of the quad at columns c+2p, c+2p+1).
+__device__ __forceinline__ void quad_transpose(const bf16x8& v, int st_r, int out[4]) {
+#pragma unroll
+  for (int k = 0; k < 4; ++k) out[k] = reinterpret_cast<const int*>(&v)[k];
+  {
+    int t0 = __shfl_xor(out[(st_r & 1) ^ 1], 16);
+    int t1 = __shfl_xor(out[((st_r & 1) ^ 1) | 2], 16);
+    if (st_r & 1) { out[0] = t0; out[2] = t1; } else { out[1] = t0; out[3] = t1; }
+  }
+  {
+    int lo = (st_r & 2) ? 0 : 2;
+    int t0 = __shfl_xor(out[lo], 32);
+    int t1 = __shfl_xor(out[lo + 1], 32);
+    out[lo] = t0; out[lo + 1] = t1;
+  }
+}
+
+// write the quad-transposed columns into a [128][stride] image (elem units);
+// r0 = quad's first row, d0 = first column this lane owns (c + 2*st_r).
+__device__ __forceinline__ void write_transposed(char* img, int stride, int r0, int d0,
+                                                 const int dw[4]) {
+#pragma unroll
+  for (int e = 0; e < 2; ++e) {
+    const int d = d0 + e;
+    unsigned w01 = e ? (((unsigned)dw[0] >> 16) | ((unsigned)dw[1] & 0xffff0000u))
+                     : (((unsigned)dw[0] & 0xffffu) | ((unsigned)dw[1] << 16));
+    unsigned w23 = e ? (((unsigned)dw[2] >> 16) | ((unsigned)dw[3] & 0xffff0000u))
+                     : (((unsigned)dw[2] & 0xffffu) | ((unsigned)dw[3] << 16));
+    int2v pair = {(int)w01, (int)w23};
+    *reinterpret_cast<int2v*>(img + (d * stride + r0) * 2) = pair;
+  }
+}
+
+// pack one 16-reg crow-ordered f32 group (this lane + its ^32 partner) into
+// two MFMA A-fragments covering k = 0..31 of that group's axis.
+__device__ __forceinline__ void pack_pair(const float p[16], bf16x8 out[2]) {
+#pragma unroll
+  for (int ksl = 0; ksl < 2; ++ksl) {
+    const float* pr = &p[8 * ksl];
+    const unsigned x1 = cvt_pk_bf16_(pr[0], pr[1]);
+    const unsigned x2 = cvt_pk_bf16_(pr[2], pr[3]);
+    const unsigned y1 = cvt_pk_bf16_(pr[4], pr[5]);
+    const unsigned y2 = cvt_pk_bf16_(pr[6], pr[7]);
+    const int2v a = __builtin_amdgcn_permlane32_swap((int)x1, (int)y1, false, false);
+    const int2v b = __builtin_amdgcn_permlane32_swap((int)x2, (int)y2, false, false);
+    int w[4] = {a[0], b[0], a[1], b[1]};
+    out[ksl] = *reinterpret_cast<const bf16x8*>(w);
+  }
 }
 
 // ---------------------------------------------------------------------------
 // delta = rowsum(dO * O), written as [B,H,S] fp32 (same layout as LSE2).
-// One 16-lane group per (b,s,h) row: 16 lanes x 8 elems = 128 = D.
 __global__ __launch_bounds__(256) void attn_bwd_delta_kernel(
     const short* __restrict__ dO, const short* __restrict__ O,
     float* __restrict__ delta, int64_t rows, int S, int H) {
@@ -58,16 +126,10 @@ __global__ __launch_bounds__(256) void attn_bwd_delta_kernel(
     const bf16x8 o8 = *reinterpret_cast<const bf16x8*>(&O[r * AB_D + lg * 8]);
     float acc = 0.f;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      union { unsigned u; float f; } a, b;
-      a.u = ((unsigned)(unsigned short)d8[j]) << 16;
-      b.u = ((unsigned)(unsigned short)o8[j]) << 16;
-      acc += a.f * b.f;
-    }
+    for (int j = 0; j < 8; ++j) acc += bf2f_(d8[j]) * bf2f_(o8[j]);
 #pragma unroll
     for (int m = 1; m < 16; m <<= 1) acc += __shfl_xor(acc, m);
     if (lg == 0) {
-      // r = (b*S + s)*H + h  ->  delta[(b*H + h)*S + s]
       const int h = (int)(r % H);
       const int64_t bs = r / H;
       const int s = (int)(bs % S);
@@ -78,357 +140,330 @@ __global__ __launch_bounds__(256) void attn_bwd_delta_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// dK/dV: workgroup = 4 waves x 16 kv rows = 64 kv rows; grid (S/64, B*HKV).
-// Loops q tiles of 32 rows from the causal diagonal to S, and (for GQA) over
-// the G query heads sharing this kv head, accumulating dK/dV in registers.
-constexpr int KV_PER_WAVE = 16;
-constexpr int KV_WG = 64;
-constexpr int QT = 32;          // q rows per staged tile
-constexpr int TP = 40;          // transposed-tile row stride (32 + 8 pad)
+// dQ: 8 waves x 32 q rows (lane&31 = own q row), KV tiles of 64,
+// double-buffered K (normal), V (normal), K^T (stride-72 image).
+constexpr int DQ_QW = 32, DQ_WAVES = 8, DQ_QB = 256, DQ_KVB = 64;
+constexpr int KT_STRIDE = 72;  // odd word stride -> conflict-free B reads
 
-__global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
-    const short* __restrict__ Q, const short* __restrict__ K,
-    const short* __restrict__ V, const short* __restrict__ dO,
-    const float* __restrict__ LSE2, const float* __restrict__ Delta,
-    short* __restrict__ dK, short* __restrict__ dV,
-    int B, int S, int H, int HKV, float c, float scale) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* q_lds = reinterpret_cast<short*>(smem);                 // [32][128] swizzled
-  short* qt_lds = q_lds + QT * AB_D;                             // [128][40]
-  short* do_lds = qt_lds + AB_D * TP;                            // [32][128] swizzled
-  short* dot_lds = do_lds + QT * AB_D;                           // [128][40]
-  short* p_lds = dot_lds + AB_D * TP;                            // per wave [16][40]
-
-  const int tid = threadIdx.x;
-  const int wid = tid >> 6;
-  const int lane = tid & 63;
-  const int lg = lane & 15;
-  const int hi = lane >> 4;
-
-  const int b = blockIdx.y / HKV;
-  const int hkv = blockIdx.y % HKV;
-  const int G = H / HKV;
-  const int kw0 = blockIdx.x * KV_WG + wid * KV_PER_WAVE;  // this wave's kv rows
-
-  const int64_t sHD = (int64_t)H * AB_D;
-  const int64_t sHkvD = (int64_t)HKV * AB_D;
-  const int64_t kv_base = (((int64_t)b * S) * HKV + hkv) * AB_D;
-
-  // K/V fragments for this wave's 16 kv rows (A-frag: lane row lg, k chunks)
-  bf16x8 kf[4], vf[4];
-  {
-    const int row = min(kw0 + lg, S - 1);
-    const int64_t rb = kv_base + (int64_t)row * sHkvD;
-#pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
-      kf[kk] = *reinterpret_cast<const bf16x8*>(&K[rb + kk * 32 + hi * 8]);
-      vf[kk] = *reinterpret_cast<const bf16x8*>(&V[rb + kk * 32 + hi * 8]);
-    }
-  }
-
-  float dv_acc[8][4], dk_acc[8][4];
-#pragma unroll
-  for (int dt = 0; dt < 8; ++dt)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) { dv_acc[dt][r] = 0.f; dk_acc[dt][r] = 0.f; }
-
-  short* pw = p_lds + wid * KV_PER_WAVE * TP;
-
-  for (int g = 0; g < G; ++g) {
-    const int h = hkv * G + g;
-    const int64_t q_base = (((int64_t)b * S) * H + h) * AB_D;
-    const int64_t ld_base = ((int64_t)b * H + h) * S;
-
-    for (int qt0 = blockIdx.x * KV_WG; qt0 < S; qt0 += QT) {
-      // ---- cooperative staging of Q and dO tiles (normal + transposed) ----
-      __syncthreads();
-#pragma unroll
-      for (int pass = 0; pass < 2; ++pass) {
-        const int i = (tid + pass * 256) * 8;
-        const int r = i / AB_D, ccol = i % AB_D;
-        const int64_t rb = q_base + (int64_t)min(qt0 + r, S - 1) * sHD + ccol;
-        const bf16x8 q8 = *reinterpret_cast<const bf16x8*>(&Q[rb]);
-        const bf16x8 d8 = *reinterpret_cast<const bf16x8*>(&dO[rb]);
-        *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(q_lds) + bswz(r, ccol)) = q8;
-        *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(do_lds) + bswz(r, ccol)) = d8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          qt_lds[(ccol + j) * TP + r] = q8[j];
-          dot_lds[(ccol + j) * TP + r] = d8[j];
-        }
-      }
-      __syncthreads();
-
-      if (kw0 > qt0 + QT - 1) continue;  // causal: no q in tile reaches this wave
-
-      // ---- S^T = K Q^T over the 32 q cols (2 n-tiles) ----
-      f32x4 st[2];
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt) st[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-#pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-          const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(q_lds) + bswz(nt * 16 + lg, kk * 32 + hi * 8));
-          st[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[kk], bq, st[nt], 0, 0, 0);
-        }
-      }
-
-      // per-lane L/delta for q col (n = nt*16 + lg)
-      float Lq[2], Dq[2];
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt) {
-        const int qcol = min(qt0 + nt * 16 + lg, S - 1);
-        Lq[nt] = LSE2[ld_base + qcol];
-        Dq[nt] = Delta[ld_base + qcol];
-      }
-
-      // ---- P^T = exp2(c*S^T - L[q]), causal+bounds masked ----
-      float p[2][4];
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt) {
-        const int qcol = qt0 + nt * 16 + lg;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kvrow = kw0 + hi * 4 + r;
-          const bool ok = (qcol >= kvrow) && (qcol < S) && (kvrow < S);
-          p[nt][r] = ok ? exp2f(st[nt][r] * c - Lq[nt]) : 0.f;
-        }
-      }
-
-      // route P^T through per-wave LDS -> A-frags
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          pw[(hi * 4 + r) * TP + nt * 16 + lg] = f2bf_(p[nt][r]);
-      const bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pw[lg * TP + hi * 8]);
-
-      // ---- dV += P^T dO  (B from transposed dO tile) ----
-#pragma unroll
-      for (int dt = 0; dt < 8; ++dt) {
-        const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
-            &dot_lds[(dt * 16 + lg) * TP + hi * 8]);
-        f32x4 acc = {dv_acc[dt][0], dv_acc[dt][1], dv_acc[dt][2], dv_acc[dt][3]};
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, bd, acc, 0, 0, 0);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) dv_acc[dt][r] = acc[r];
-      }
-
-      // ---- dP^T = V dO^T ----
-      f32x4 dpt[2];
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt) dpt[nt] = f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-#pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-          const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(do_lds) + bswz(nt * 16 + lg, kk * 32 + hi * 8));
-          dpt[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[kk], bd, dpt[nt], 0, 0, 0);
-        }
-      }
-
-      // ---- dS^T = P^T * (dP^T - delta[q]) -> LDS -> A-frags ----
-#pragma unroll
-      for (int nt = 0; nt < 2; ++nt)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          pw[(hi * 4 + r) * TP + nt * 16 + lg] = f2bf_(p[nt][r] * (dpt[nt][r] - Dq[nt]));
-      const bf16x8 dsf = *reinterpret_cast<const bf16x8*>(&pw[lg * TP + hi * 8]);
-
-      // ---- dK += dS^T Q  (B from transposed Q tile) ----
-#pragma unroll
-      for (int dt = 0; dt < 8; ++dt) {
-        const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-            &qt_lds[(dt * 16 + lg) * TP + hi * 8]);
-        f32x4 acc = {dk_acc[dt][0], dk_acc[dt][1], dk_acc[dt][2], dk_acc[dt][3]};
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, bq, acc, 0, 0, 0);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) dk_acc[dt][r] = acc[r];
-      }
-    }
-  }
-
-  // ---- epilogue: store this wave's 16 kv rows ----
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = kw0 + hi * 4 + r;
-    if (row >= S) continue;
-    const int64_t rb = kv_base + (int64_t)row * sHkvD;
-#pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
-      dV[rb + dt * 16 + lg] = f2bf_(dv_acc[dt][r]);
-      dK[rb + dt * 16 + lg] = f2bf_(dk_acc[dt][r] * scale);
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// dQ: workgroup = 4 waves x 32 q rows = 128 q rows; grid (S/128, B*H).
-// Loops kv tiles of 32 rows from 0 to the causal diagonal.
-constexpr int QW = 32;   // q rows per wave
-constexpr int QWG = 128;
-constexpr int KT = 32;   // kv rows per staged tile
-
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
     const float* __restrict__ LSE2, const float* __restrict__ Delta,
     short* __restrict__ dQ,
     int B, int S, int H, int HKV, float c, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = reinterpret_cast<short*>(smem);            // [32][128] swizzled
-  short* kt_lds = k_lds + KT * AB_D;                        // [128][40]
-  short* v_lds = kt_lds + AB_D * TP;                        // [32][128] swizzled
-  short* a_lds = v_lds + KT * AB_D;                         // per wave [32][40]
+  // K0@0 K1@16K V0@32K V1@48K Kt0@64K Kt1@64K+18432
+  auto k_lds = [&](int buf) -> char* { return smem + buf * 16384; };
+  auto v_lds = [&](int buf) -> char* { return smem + 32768 + buf * 16384; };
+  auto kt_lds = [&](int buf) -> char* { return smem + 65536 + buf * (AB_D * KT_STRIDE * 2); };
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int lg = lane & 15;
-  const int hi = lane >> 4;
+  const int lq = lane & 31;
+  const int hi2 = lane >> 5;
 
   const int b = blockIdx.y / H;
   const int h = blockIdx.y % H;
   const int hkv = h / (H / HKV);
-  const int qw0 = blockIdx.x * QWG + wid * QW;
-
   const int64_t sHD = (int64_t)H * AB_D;
   const int64_t sHkvD = (int64_t)HKV * AB_D;
   const int64_t q_base = (((int64_t)b * S) * H + h) * AB_D;
   const int64_t kv_base = (((int64_t)b * S) * HKV + hkv) * AB_D;
   const int64_t ld_base = ((int64_t)b * H + h) * S;
 
-  // Q and dO fragments: [2 m-tiles][4 k-chunks]
-  bf16x8 qf[2][4], dof[2][4];
+  const int qw0 = blockIdx.x * DQ_QB + wid * DQ_QW;
+  const int qrow = qw0 + lq;
+
+  bf16x8 qf[8], dof[8];
+  {
+    const int64_t rb = q_base + (int64_t)min(qrow, S - 1) * sHD;
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi) {
-    const int row = min(qw0 + qi * 16 + lg, S - 1);
+    for (int dc = 0; dc < 8; ++dc) {
+      qf[dc] = *reinterpret_cast<const bf16x8*>(&Q[rb + dc * 16 + hi2 * 8]);
+      dof[dc] = *reinterpret_cast<const bf16x8*>(&dO[rb + dc * 16 + hi2 * 8]);
+    }
+  }
+  const float Lq = LSE2[ld_base + min(qrow, S - 1)];
+  const float Dq = Delta[ld_base + min(qrow, S - 1)];
+
+  const int st_r = lane >> 4;
+  const int st_c = 8 * (lane & 15);
+  const int kv_end = min(S, (int)(blockIdx.x + 1) * DQ_QB);
+
+  auto ld_tile = [&](int kv0, int pass, bf16x8& kreg, bf16x8& vreg) {
+    const int row = min(kv0 + 4 * wid + 32 * pass + st_r, S - 1);
+    const int64_t rb = kv_base + (int64_t)row * sHkvD + st_c;
+    kreg = *reinterpret_cast<const bf16x8*>(&K[rb]);
+    vreg = *reinterpret_cast<const bf16x8*>(&V[rb]);
+  };
+  auto write_tile = [&](int buf, int pass, bf16x8 kreg, bf16x8 vreg) {
+    const int r = 4 * wid + 32 * pass + st_r;
+    *reinterpret_cast<bf16x8*>(k_lds(buf) + bswz(r, st_c)) = kreg;
+    *reinterpret_cast<bf16x8*>(v_lds(buf) + bswz(r, st_c)) = vreg;
+    int dw[4];
+    quad_transpose(kreg, st_r, dw);
+    write_transposed(kt_lds(buf), KT_STRIDE, 4 * wid + 32 * pass, st_c + 2 * st_r, dw);
+  };
+
+  f32x16 dq_acc[4];
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
+
+  bf16x8 kreg0, kreg1, vreg0, vreg1;
+  ld_tile(0, 0, kreg0, vreg0);
+  ld_tile(0, 1, kreg1, vreg1);
+  write_tile(0, 0, kreg0, vreg0);
+  write_tile(0, 1, kreg1, vreg1);
+  __syncthreads();
+
+  for (int kv0 = 0, cur = 0; kv0 < kv_end; kv0 += DQ_KVB, cur ^= 1) {
+    const bool have_next = kv0 + DQ_KVB < kv_end;
+    if (have_next) {
+      ld_tile(kv0 + DQ_KVB, 0, kreg0, vreg0);
+      ld_tile(kv0 + DQ_KVB, 1, kreg1, vreg1);
+    }
+
+    if (kv0 < qw0 + DQ_QW) {
+      f32x16 st[2], dpt[2];
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { st[t2][r] = 0.f; dpt[t2][r] = 0.f; }
+#pragma unroll
+      for (int dc = 0; dc < 8; ++dc) {
+#pragma unroll
+        for (int t2 = 0; t2 < 2; ++t2) {
+          const bf16x8 ak = *reinterpret_cast<const bf16x8*>(
+              k_lds(cur) + bswz(t2 * 32 + lq, dc * 16 + hi2 * 8));
+          const bf16x8 av = *reinterpret_cast<const bf16x8*>(
+              v_lds(cur) + bswz(t2 * 32 + lq, dc * 16 + hi2 * 8));
+          st[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[dc], st[t2], 0, 0, 0);
+          dpt[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[dc], dpt[t2], 0, 0, 0);
+        }
+      }
+
+      const bool need_mask = (kv0 + DQ_KVB > qw0) || (kv0 + DQ_KVB > S);
+      float ds[32];
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          bool ok = true;
+          if (need_mask) {
+            const int kv = kv0 + t2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi2;
+            ok = (kv <= qrow) && (kv < S);
+          }
+          const float p = ok ? exp2f(fmaf(st[t2][r], c, -Lq)) : 0.f;
+          ds[t2 * 16 + r] = p * (dpt[t2][r] - Dq);
+        }
+
+      bf16x8 dsa[4];
+      pack_pair(&ds[0], &dsa[0]);
+      pack_pair(&ds[16], &dsa[2]);
+
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const int d = dt * 32 + lq;
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
+              kt_lds(cur) + (d * KT_STRIDE + ks * 16 + hi2 * 8) * 2);
+          dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[ks], bk, dq_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+
+    if (have_next) {
+      write_tile(cur ^ 1, 0, kreg0, vreg0);
+      write_tile(cur ^ 1, 1, kreg1, vreg1);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = qw0 + (r & 3) + 8 * (r >> 2) + 4 * hi2;
+    if (row >= S) continue;
     const int64_t rb = q_base + (int64_t)row * sHD;
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
-      qf[qi][kk] = *reinterpret_cast<const bf16x8*>(&Q[rb + kk * 32 + hi * 8]);
-      dof[qi][kk] = *reinterpret_cast<const bf16x8*>(&dO[rb + kk * 32 + hi * 8]);
-    }
+    for (int dt = 0; dt < 4; ++dt) dQ[rb + dt * 32 + lq] = f2bf_(dq_acc[dt][r] * scale);
   }
-  // per-C-row L/delta
-  float Lr[2][4], Dr[2][4];
-#pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int row = min(qw0 + qi * 16 + hi * 4 + r, S - 1);
-      Lr[qi][r] = LSE2[ld_base + row];
-      Dr[qi][r] = Delta[ld_base + row];
-    }
+}
 
-  float dq_acc[2][8][4];
-#pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-    for (int dt = 0; dt < 8; ++dt)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) dq_acc[qi][dt][r] = 0.f;
+// ---------------------------------------------------------------------------
+// dK/dV: 4 waves x 32 kv rows (lane&31 = own kv row), 256 threads,
+// 1 wave/SIMD (full 512-VGPR budget), q tiles of 32,
+// double-buffered Q (normal), dO (normal), Q^T, dO^T (stride-40 images),
+// L/Delta staged per tile. GQA: accumulate over the group's query heads.
+constexpr int KV_KW = 32, KV_WAVES = 4, KV_WG = 128, KV_QT = 32;
+constexpr int QT_STRIDE = 40;
 
-  short* aw = a_lds + wid * QW * TP;
-  const int kv_end = min(S, blockIdx.x * QWG + QWG);
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkdv_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE2, const float* __restrict__ Delta,
+    short* __restrict__ dK, short* __restrict__ dV,
+    int B, int S, int H, int HKV, float c, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // Q0@0 Q1@8K dO0@16K dO1@24K Qt0@32768 Qt1@+10240 dOt0@53248 dOt1@+10240
+  // L@73728 (2 x 32 f32) D@74240
+  auto q_lds = [&](int buf) -> char* { return smem + buf * 8192; };
+  auto do_lds = [&](int buf) -> char* { return smem + 16384 + buf * 8192; };
+  auto qt_lds = [&](int buf) -> char* { return smem + 32768 + buf * (AB_D * QT_STRIDE * 2); };
+  auto dot_lds = [&](int buf) -> char* { return smem + 53248 + buf * (AB_D * QT_STRIDE * 2); };
+  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 73728) + buf * 32; };
+  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 74240) + buf * 32; };
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KT) {
-    // ---- stage K (normal swizzled + transposed) and V (normal swizzled) ----
-    __syncthreads();
-#pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int i = (tid + pass * 256) * 8;
-      const int r = i / AB_D, ccol = i % AB_D;
-      const int64_t rb = kv_base + (int64_t)min(kv0 + r, S - 1) * sHkvD + ccol;
-      const bf16x8 k8 = *reinterpret_cast<const bf16x8*>(&K[rb]);
-      const bf16x8 v8 = *reinterpret_cast<const bf16x8*>(&V[rb]);
-      *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(k_lds) + bswz(r, ccol)) = k8;
-      *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(v_lds) + bswz(r, ccol)) = v8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) kt_lds[(ccol + j) * TP + r] = k8[j];
-    }
-    __syncthreads();
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int lq = lane & 31;
+  const int hi2 = lane >> 5;
 
-    if (kv0 >= qw0 + QW) continue;  // causal: kv tile entirely above this wave
+  const int b = blockIdx.y / HKV;
+  const int hkv = blockIdx.y % HKV;
+  const int G = H / HKV;
+  const int64_t sHD = (int64_t)H * AB_D;
+  const int64_t sHkvD = (int64_t)HKV * AB_D;
+  const int64_t kv_base = (((int64_t)b * S) * HKV + hkv) * AB_D;
 
-    // ---- S = Q K^T ----
-    f32x4 s_acc[2][2], dp_acc[2][2];
-#pragma unroll
-    for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-      for (int ki = 0; ki < 2; ++ki) {
-        s_acc[qi][ki] = f32x4{0.f, 0.f, 0.f, 0.f};
-        dp_acc[qi][ki] = f32x4{0.f, 0.f, 0.f, 0.f};
-      }
-#pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
-#pragma unroll
-      for (int ki = 0; ki < 2; ++ki) {
-        const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<char*>(k_lds) + bswz(ki * 16 + lg, kk * 32 + hi * 8));
-        const bf16x8 bv = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<char*>(v_lds) + bswz(ki * 16 + lg, kk * 32 + hi * 8));
-#pragma unroll
-        for (int qi = 0; qi < 2; ++qi) {
-          s_acc[qi][ki] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[qi][kk], bk, s_acc[qi][ki], 0, 0, 0);
-          dp_acc[qi][ki] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[qi][kk], bv, dp_acc[qi][ki], 0, 0, 0);
-        }
-      }
-    }
+  const int kvw0 = blockIdx.x * KV_WG + wid * KV_KW;  // wave's kv rows
+  const int kvrow = kvw0 + lq;                        // lane's kv row
 
-    // ---- dS = P * (dP - delta) -> per-wave LDS ----
+  // K/V B-fragments for this wave's 32 kv rows (lane n = kv row)
+  bf16x8 kf[8], vf[8];
+  {
+    const int64_t rb = kv_base + (int64_t)min(kvrow, S - 1) * sHkvD;
 #pragma unroll
-    for (int qi = 0; qi < 2; ++qi)
-#pragma unroll
-      for (int ki = 0; ki < 2; ++ki)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = qw0 + qi * 16 + hi * 4 + r;
-          const int col = kv0 + ki * 16 + lg;
-          float pv = 0.f;
-          if (col <= row && col < S)
-            pv = exp2f(s_acc[qi][ki][r] * c - Lr[qi][r]);
-          aw[(qi * 16 + hi * 4 + r) * TP + ki * 16 + lg] =
-              f2bf_(pv * (dp_acc[qi][ki][r] - Dr[qi][r]));
-        }
-
-    // ---- dQ += dS K  (A from per-wave LDS, B from transposed K tile) ----
-    bf16x8 af[2];
-#pragma unroll
-    for (int qi = 0; qi < 2; ++qi)
-      af[qi] = *reinterpret_cast<const bf16x8*>(&aw[(qi * 16 + lg) * TP + hi * 8]);
-#pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
-      const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
-          &kt_lds[(dt * 16 + lg) * TP + hi * 8]);
-#pragma unroll
-      for (int qi = 0; qi < 2; ++qi) {
-        f32x4 acc = {dq_acc[qi][dt][0], dq_acc[qi][dt][1], dq_acc[qi][dt][2],
-                     dq_acc[qi][dt][3]};
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[qi], bk, acc, 0, 0, 0);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) dq_acc[qi][dt][r] = acc[r];
-      }
+    for (int dc = 0; dc < 8; ++dc) {
+      kf[dc] = *reinterpret_cast<const bf16x8*>(&K[rb + dc * 16 + hi2 * 8]);
+      vf[dc] = *reinterpret_cast<const bf16x8*>(&V[rb + dc * 16 + hi2 * 8]);
     }
   }
 
-  // ---- epilogue ----
+  f32x16 dv_acc[4], dk_acc[4];
 #pragma unroll
-  for (int qi = 0; qi < 2; ++qi)
+  for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int row = qw0 + qi * 16 + hi * 4 + r;
-      if (row >= S) continue;
-      const int64_t rb = q_base + (int64_t)row * sHD;
+    for (int r = 0; r < 16; ++r) { dv_acc[dt][r] = 0.f; dk_acc[dt][r] = 0.f; }
+
+  const int st_r = lane >> 4;
+  const int st_c = 8 * (lane & 15);
+  const int qstart = blockIdx.x * KV_WG;  // first q tile with any work
+
+  for (int g = 0; g < G; ++g) {
+    const int h = hkv * G + g;
+    const int64_t q_base = (((int64_t)b * S) * H + h) * AB_D;
+    const int64_t ld_base = ((int64_t)b * H + h) * S;
+
+    auto ld_tile = [&](int qt0, int pass, bf16x8& qreg, bf16x8& dreg) {
+      const int row = min(qt0 + 4 * wid + 16 * pass + st_r, S - 1);
+      const int64_t rb = q_base + (int64_t)row * sHD + st_c;
+      qreg = *reinterpret_cast<const bf16x8*>(&Q[rb]);
+      dreg = *reinterpret_cast<const bf16x8*>(&dO[rb]);
+    };
+    auto write_tile = [&](int buf, int qt0, int pass, bf16x8 qreg, bf16x8 dreg) {
+      const int r = 4 * wid + 16 * pass + st_r;
+      *reinterpret_cast<bf16x8*>(q_lds(buf) + bswz(r, st_c)) = qreg;
+      *reinterpret_cast<bf16x8*>(do_lds(buf) + bswz(r, st_c)) = dreg;
+      int dw[4];
+      quad_transpose(qreg, st_r, dw);
+      write_transposed(qt_lds(buf), QT_STRIDE, 4 * wid + 16 * pass, st_c + 2 * st_r, dw);
+      quad_transpose(dreg, st_r, dw);
+      write_transposed(dot_lds(buf), QT_STRIDE, 4 * wid + 16 * pass, st_c + 2 * st_r, dw);
+      if (pass == 0 && tid < 64) {  // L/Delta for the tile's 32 q rows
+        const int qi = min(qt0 + (tid & 31), S - 1);
+        if (tid < 32) l_buf(buf)[tid] = LSE2[ld_base + qi];
+        else d_buf(buf)[tid & 31] = Delta[ld_base + qi];
+      }
+    };
+
+    bf16x8 qreg0, qreg1, dreg0, dreg1;
+    ld_tile(qstart, 0, qreg0, dreg0);
+    ld_tile(qstart, 1, qreg1, dreg1);
+    write_tile(0, qstart, 0, qreg0, dreg0);
+    write_tile(0, qstart, 1, qreg1, dreg1);
+    __syncthreads();
+
+    for (int qt0 = qstart, cur = 0; qt0 < S; qt0 += KV_QT, cur ^= 1) {
+      const bool have_next = qt0 + KV_QT < S;
+      if (have_next) {
+        ld_tile(qt0 + KV_QT, 0, qreg0, dreg0);
+        ld_tile(qt0 + KV_QT, 1, qreg1, dreg1);
+      }
+
+      if (qt0 + KV_QT - 1 >= kvw0) {  // causal: some q in tile reaches this wave
+        f32x16 st, dpt;
 #pragma unroll
-      for (int dt = 0; dt < 8; ++dt)
-        dQ[rb + dt * 16 + lg] = f2bf_(dq_acc[qi][dt][r] * scale);
+        for (int r = 0; r < 16; ++r) { st[r] = 0.f; dpt[r] = 0.f; }
+#pragma unroll
+        for (int dc = 0; dc < 8; ++dc) {
+          const bf16x8 aq = *reinterpret_cast<const bf16x8*>(
+              q_lds(cur) + bswz(lq, dc * 16 + hi2 * 8));
+          const bf16x8 ad = *reinterpret_cast<const bf16x8*>(
+              do_lds(cur) + bswz(lq, dc * 16 + hi2 * 8));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[dc], st, 0, 0, 0);
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[dc], dpt, 0, 0, 0);
+        }
+
+        const bool need_mask = (qt0 < kvw0 + KV_KW) || (qt0 + KV_QT > S);
+        float p[16], ds[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi2;  // q row of reg r
+          const float Lr = l_buf(cur)[qr];
+          const float Dr = d_buf(cur)[qr];
+          bool ok = (kvrow < S);
+          if (need_mask) {
+            const int q = qt0 + qr;
+            ok = ok && (q >= kvrow) && (q < S);
+          }
+          p[r] = ok ? exp2f(fmaf(st[r], c, -Lr)) : 0.f;
+          ds[r] = p[r] * (dpt[r] - Dr);
+        }
+
+        bf16x8 pa[2], dsa[2];
+        pack_pair(p, pa);
+        pack_pair(ds, dsa);
+
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          const int d = dt * 32 + lq;
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
+                dot_lds(cur) + (d * QT_STRIDE + ks * 16 + hi2 * 8) * 2);
+            const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+                qt_lds(cur) + (d * QT_STRIDE + ks * 16 + hi2 * 8) * 2);
+            dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], bd, dv_acc[dt], 0, 0, 0);
+            dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[ks], bq, dk_acc[dt], 0, 0, 0);
+          }
+        }
+      }
+
+      if (have_next) {
+        write_tile(cur ^ 1, qt0 + KV_QT, 0, qreg0, dreg0);
+        write_tile(cur ^ 1, qt0 + KV_QT, 1, qreg1, dreg1);
+      }
+      __syncthreads();
     }
+    __syncthreads();  // buffer 0 reuse across g
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = kvw0 + (r & 3) + 8 * (r >> 2) + 4 * hi2;
+    if (row >= S) continue;
+    const int64_t rb = kv_base + (int64_t)row * sHkvD;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      dV[rb + dt * 32 + lq] = f2bf_(dv_acc[dt][r]);
+      dK[rb + dt * 32 + lq] = f2bf_(dk_acc[dt][r] * scale);
+    }
+  }
 }
 
 }  // namespace lpp
@@ -462,28 +497,24 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   auto dv = at::empty_like(v);
 
   {
+    const int qblocks = (S + lpp::DQ_QB - 1) / lpp::DQ_QB;
+    const size_t lds = 65536 + 2 * (lpp::AB_D * lpp::KT_STRIDE * 2);
+    hipLaunchKernelGGL(lpp::attn_bwd_dq_kernel, dim3(qblocks, B * H), dim3(512), lds,
+                       stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                       (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
+                       lse2.data_ptr<float>(), delta.data_ptr<float>(),
+                       (short*)dq.data_ptr(), B, S, H, HKV, c, scale);
+    LPP_CHECK_HIP(hipGetLastError());
+  }
+  {
     const int kvblocks = (S + lpp::KV_WG - 1) / lpp::KV_WG;
-    const size_t lds = (2 * (lpp::QT * lpp::AB_D + lpp::AB_D * lpp::TP) +
-                        4 * lpp::KV_PER_WAVE * lpp::TP) *
-                       sizeof(short);
+    const size_t lds = 74496;
     hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(256), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
                        lse2.data_ptr<float>(), delta.data_ptr<float>(),
                        (short*)dk.data_ptr(), (short*)dv.data_ptr(), B, S, H, HKV, c,
                        scale);
-    LPP_CHECK_HIP(hipGetLastError());
-  }
-  {
-    const int qblocks = (S + lpp::QWG - 1) / lpp::QWG;
-    const size_t lds = (2 * lpp::KT * lpp::AB_D + lpp::AB_D * lpp::TP +
-                        4 * lpp::QW * lpp::TP) *
-                       sizeof(short);
-    hipLaunchKernelGGL(lpp::attn_bwd_dq_kernel, dim3(qblocks, B * H), dim3(256), lds,
-                       stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                       (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
-                       lse2.data_ptr<float>(), delta.data_ptr<float>(),
-                       (short*)dq.data_ptr(), B, S, H, HKV, c, scale);
     LPP_CHECK_HIP(hipGetLastError());
   }
   return {dq, dk, dv};
