@@ -1,0 +1,141 @@
+"""Text data pipeline vs the reference's data-layer behavior (CPU).
+
+Covers the SURVEY.md §2.8 inventory: seq2seq->causal-LM conversion with
+completion-only labels (data/flan.py:149-190), the collator contract
+(:263-309 with quirks Q1/Q2 fixed), dataset mixing (:65-146), the
+placeholder dataset (data/test.py), and special-token expansion
+(general_util/tokenization_utils.py:15-56).
+"""
+
+import json
+import os
+
+import pytest
+import torch
+
+from lpp_amd.data import (
+    IGNORE_INDEX,
+    FlattenedGroupDataset,
+    PlaceholderDataset,
+    PromptResponseDataset,
+    RoundRobinMixDataset,
+    SimpleTokenizer,
+    TextCollator,
+    completion_labels,
+    expand_special_tokenizer,
+)
+
+
+def test_expand_special_tokenizer_defaults():
+    tok = SimpleTokenizer()
+    assert tok.pad_token is None
+    added = expand_special_tokenizer(tok)
+    assert added == 1 and tok.pad_token == "[PAD]"
+    assert tok.pad_token_id is not None
+
+
+def test_expand_special_tokenizer_pad_falls_back_to_eos():
+    tok = SimpleTokenizer()
+    tok.name_or_path = "gpt-something"  # non-llama path
+    expand_special_tokenizer(tok)
+    assert tok.pad_token == tok.eos_token
+    assert tok.pad_token_id == tok.eos_token_id
+
+
+def test_completion_labels_masks_prompt_and_pad():
+    ids = torch.tensor([[5, 6, 7, 8, 0, 0], [9, 10, 11, 12, 13, 14]])
+    lens = torch.tensor([2, 3])
+    labels = completion_labels(ids, lens, pad_token_id=0)
+    assert labels[0].tolist() == [-100, -100, 7, 8, -100, -100]
+    assert labels[1].tolist() == [-100, -100, -100, 12, 13, 14]
+
+
+def test_text_collator_contract():
+    tok = SimpleTokenizer()
+    coll = TextCollator(tok, max_seq_length=16)
+    batch = [
+        {"inputs": "what is two plus two ?", "targets": "four"},
+        {"inputs": "name a color", "targets": "red or blue"},
+    ]
+    out = coll(batch)
+    # the engine contract: exactly input_ids + labels, same [B, S] shape
+    assert set(out) == {"input_ids", "labels"}
+    assert out["input_ids"].shape == (2, 16)
+    assert out["labels"].shape == (2, 16)
+    # loss only on completion tokens: prompt positions ignored
+    for i, ex in enumerate(batch):
+        n_prompt = len(ex["inputs"].split()) + 1  # + bos
+        assert (out["labels"][i, :n_prompt] == IGNORE_INDEX).all()
+        n_full = len((ex["inputs"] + " " + ex["targets"]).split()) + 2  # bos+eos
+        assert (out["labels"][i, n_prompt:n_full] != IGNORE_INDEX).all()
+        # pad tail ignored
+        assert (out["labels"][i, n_full:] == IGNORE_INDEX).all()
+    # labels equal input_ids where not ignored (shift happens in the loss)
+    keep = out["labels"] != IGNORE_INDEX
+    assert torch.equal(out["labels"][keep], out["input_ids"][keep])
+
+
+def test_text_collator_nested_field():
+    tok = SimpleTokenizer()
+    coll = TextCollator(tok, max_seq_length=8, field="flan")
+    out = coll([{"flan": {"inputs": "a b", "targets": "c"}}])
+    assert out["input_ids"].shape[0] == 1
+
+
+def test_round_robin_mix():
+    a = [{"inputs": f"a{i}", "targets": "x"} for i in range(3)]
+    b = [{"inputs": f"b{i}", "targets": "y"} for i in range(5)]
+    mix = RoundRobinMixDataset(main=a, flan=b)
+    assert len(mix) == 5
+    item = mix[4]
+    assert item["main"]["inputs"] == "a1"  # 4 % 3
+    assert item["flan"]["inputs"] == "b4"
+
+
+def test_flattened_group_dataset():
+    groups = [[1, 2], [3], [4, 5, 6]]
+    ds = FlattenedGroupDataset(groups)
+    assert len(ds) == 6
+    assert [ds[i] for i in range(6)] == [1, 2, 3, 4, 5, 6]
+
+
+def test_placeholder_dataset_len():
+    ds = PlaceholderDataset(pseudo_dataset_len=123)
+    assert len(ds) == 123
+    assert ds[7]["inputs"]
+    assert len(PlaceholderDataset()) == 100000000
+
+
+def test_prompt_response_dataset_jsonl(tmp_path):
+    p = tmp_path / "d.jsonl"
+    rows = [
+        {"inputs": "q1", "targets": "a1"},
+        {"inputs": "  ", "targets": "dropped"},
+        {"prompt": "q2", "response": "a2"},
+        {"inputs": "q3", "targets": ""},
+    ]
+    with open(p, "w") as f:
+        for r in rows:
+            f.write(json.dumps(r) + "\n")
+    ds = PromptResponseDataset(str(p))
+    assert len(ds) == 2
+    assert ds[1] == {"inputs": "q2", "targets": "a2"}
+
+
+def test_env_override_eos(monkeypatch):
+    monkeypatch.setenv("EOS_TOKEN", "</s>")
+    tok = SimpleTokenizer()
+    expand_special_tokenizer(tok)
+    assert tok.eos_token == "</s>"
+
+
+def test_collator_feeds_engine_loss():
+    """End to end: text batch -> loss_fn shift semantics run clean."""
+    from lpp_amd.models import loss_fn
+
+    tok = SimpleTokenizer(vocab_size=128)
+    coll = TextCollator(tok, max_seq_length=12)
+    out = coll([{"inputs": "the cat sat", "targets": "on the mat"}])
+    logits = torch.randn(1, 12, 128)
+    loss = loss_fn(logits, out["labels"])
+    assert torch.isfinite(loss)
