@@ -31,6 +31,7 @@ import torch.nn as nn
 from ..config import ModelConfig
 from ..layer_spec import LayerSpec
 from .. import ops
+from ..ops.linear import LPLinear, lp_linear
 
 
 class RMSNorm(nn.Module):
@@ -54,11 +55,10 @@ class LlamaAttention(nn.Module):
         self.hidden_size = cfg.hidden_size
         self.rope_theta = cfg.rope_theta
         self.max_seq_len = cfg.max_seq_len
-        bias = False
-        self.q_proj = nn.Linear(cfg.hidden_size, self.num_heads * self.head_dim, bias=bias)
-        self.k_proj = nn.Linear(cfg.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
-        self.v_proj = nn.Linear(cfg.hidden_size, self.num_kv_heads * self.head_dim, bias=bias)
-        self.o_proj = nn.Linear(self.num_heads * self.head_dim, cfg.hidden_size, bias=bias)
+        self.q_proj = LPLinear(cfg.hidden_size, self.num_heads * self.head_dim)
+        self.k_proj = LPLinear(cfg.hidden_size, self.num_kv_heads * self.head_dim)
+        self.v_proj = LPLinear(cfg.hidden_size, self.num_kv_heads * self.head_dim)
+        self.o_proj = LPLinear(self.num_heads * self.head_dim, cfg.hidden_size)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
@@ -77,9 +77,9 @@ class LlamaAttention(nn.Module):
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
-        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+        self.gate_proj = LPLinear(cfg.hidden_size, cfg.intermediate_size)
+        self.up_proj = LPLinear(cfg.hidden_size, cfg.intermediate_size)
+        self.down_proj = LPLinear(cfg.intermediate_size, cfg.hidden_size)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
@@ -149,6 +149,9 @@ class LMHeadPipe(nn.Linear):
 
     def __init__(self, hidden_size: int, vocab_size: int):
         super().__init__(hidden_size, vocab_size, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return lp_linear(x, self.weight)
 
     @staticmethod
     def spec_param_count(hidden_size: int, vocab_size: int) -> int:

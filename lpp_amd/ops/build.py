@@ -96,6 +96,7 @@ def build(verbose: bool = True, force: bool = False) -> Path:
         "-lc10",
         f"-L{rocm}/lib",
         "-lamdhip64",
+        "-lhipblaslt",
     ]
     # torch hip libs
     if (torch_lib / "libtorch_hip.so").exists():

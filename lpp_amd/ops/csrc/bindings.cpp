@@ -17,6 +17,7 @@ std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v);
 std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
                                       at::Tensor v, at::Tensor o, at::Tensor lse2);
 at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
+void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw);
 at::Tensor mfma_test_32x32x16(at::Tensor A, at::Tensor B);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> masters,
                  std::vector<at::Tensor> grads, std::vector<at::Tensor> exp_avg,
@@ -37,6 +38,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw, "fused mixed-precision AdamW");
   m.def("attention_fwd", &attention_fwd, "flash causal attention forward (O, LSE2)");
   m.def("attention_bwd", &attention_bwd, "flash causal attention backward (dQ, dK, dV)");
+  m.def("wgrad_f32_accum", &wgrad_f32_accum, "dW_f32 += dY^T @ X (hipBLASLt, beta=1)");
   m.def("mfma_test_16x16x32", &mfma_test_16x16x32, "MFMA layout validation");
   m.def("mfma_test_32x32x16", &mfma_test_32x32x16, "MFMA 32x32x16 layout validation");
 }
