@@ -20,6 +20,9 @@ class _LinearWgradF32(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.nn.Parameter):
         ctx.save_for_backward(x, weight)
+        # saved_tensors unwraps the Parameter, so keep the fp32 buffer
+        # reference directly (it is optimizer state, not part of the graph)
+        ctx.main_grad = weight.main_grad
         return F.linear(x, weight)
 
     @staticmethod
@@ -31,7 +34,7 @@ class _LinearWgradF32(torch.autograd.Function):
         ext.wgrad_f32_accum(
             x.reshape(-1, x.shape[-1]).contiguous(),
             dy.reshape(-1, dy.shape[-1]),
-            weight.main_grad,
+            ctx.main_grad,
         )
         return dx, None
 
