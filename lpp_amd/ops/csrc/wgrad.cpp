@@ -92,15 +92,51 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
       size_t ws = kWorkspaceBytes;
       LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceSetAttribute(
           pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-      hipblasLtMatmulHeuristicResult_t results[4];
+      hipblasLtMatmulHeuristicResult_t results[16];
       int found = 0;
       LPP_CHECK_BLASLT(hipblasLtMatmulAlgoGetHeuristic(
-          lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 4, results,
+          lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 16, results,
           &found));
       hipblasLtMatmulPreferenceDestroy(pref);
       TORCH_CHECK(found > 0, "hipblaslt: no algo for wgrad shape [", out, ",", in,
                   "] k=", T);
+      // measured pick: time each heuristic candidate once on scratch
+      // operands (first call per shape only; the heuristic's first choice
+      // was ~10% slower than the best for the 65B beta=1 f32-D shapes)
       plan.algo = results[0].algo;
+      if (found > 1) {
+        auto sx = at::empty({T, in}, x.options());
+        auto sy = at::empty({T, out}, x.options());
+        auto sw = at::zeros({out, in}, dw.options());
+        auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, x.options().dtype(at::kByte));
+        const float a1 = 1.0f, b1 = 1.0f;
+        auto stream = current_stream();
+        hipEvent_t ev0, ev1;
+        LPP_CHECK_HIP(hipEventCreate(&ev0));
+        LPP_CHECK_HIP(hipEventCreate(&ev1));
+        float best = 1e30f;
+        for (int i = 0; i < found; ++i) {
+          // one warm + one timed run per candidate
+          for (int rep = 0; rep < 2; ++rep) {
+            if (rep == 1) LPP_CHECK_HIP(hipEventRecord(ev0, stream));
+            hipblasStatus_t st = hipblasLtMatmul(
+                lt_ctx().handle, plan.op, &a1, sx.data_ptr(), plan.a, sy.data_ptr(),
+                plan.b, &b1, sw.data_ptr(), plan.c, sw.data_ptr(), plan.c,
+                &results[i].algo, wsbuf.data_ptr(), kWorkspaceBytes, stream);
+            if (st != HIPBLAS_STATUS_SUCCESS) { best = best; goto next_algo; }
+          }
+          LPP_CHECK_HIP(hipEventRecord(ev1, stream));
+          LPP_CHECK_HIP(hipEventSynchronize(ev1));
+          {
+            float ms = 0.f;
+            LPP_CHECK_HIP(hipEventElapsedTime(&ms, ev0, ev1));
+            if (ms < best) { best = ms; plan.algo = results[i].algo; }
+          }
+        next_algo:;
+        }
+        LPP_CHECK_HIP(hipEventDestroy(ev0));
+        LPP_CHECK_HIP(hipEventDestroy(ev1));
+      }
       plans.emplace(key, plan);
     }
   }
