@@ -60,7 +60,7 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--gas", type=int, default=32, help="microbatches per step")
+    ap.add_argument("--gas", type=int, default=64, help="microbatches per step")
     ap.add_argument("--micro-batch-size", type=int, default=1)
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--model", type=str, default="llama-65b")

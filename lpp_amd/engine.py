@@ -23,6 +23,8 @@ loss scaling is supported for reference parity (conf/...yaml:137-143).
 from __future__ import annotations
 
 import logging
+import os
+import threading
 import time
 from collections import deque
 from typing import Iterator, Optional
@@ -65,6 +67,57 @@ class DynamicLossScaler:
                 self._good_steps = 0
 
 
+class _Watchdog:
+    """Deadlock watchdog (the reference's only backstop is
+    the 7200s NCCL timeout).  While armed, if a train_batch exceeds the
+    timeout the thread dumps this rank's schedule position to stderr so a
+    hung pipeline is diagnosable per rank instead of dying silently.
+    Enable with LPP_WATCHDOG_S=<seconds> (or TrainConfig.watchdog_timeout_s)."""
+
+    def __init__(self, engine: "PipelineEngine", timeout_s: float):
+        self.engine = engine
+        self.timeout_s = timeout_s
+        self._deadline = None
+        self._fired = False
+        self._stop = False
+        self._cv = threading.Condition()
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+
+    def arm(self):
+        with self._cv:
+            self._deadline = time.time() + self.timeout_s
+            self._fired = False
+            self._cv.notify()
+
+    def disarm(self):
+        with self._cv:
+            self._deadline = None
+
+    def _run(self):
+        while True:
+            with self._cv:
+                if self._stop:
+                    return
+                if self._deadline is None:
+                    self._cv.wait(timeout=1.0)
+                    continue
+                remaining = self._deadline - time.time()
+            if remaining <= 0:
+                if not self._fired:
+                    self._fired = True
+                    e = self.engine
+                    logger.error(
+                        "[watchdog] rank %d stage %d: train_batch stuck >%.0fs at %r "
+                        "(step %d, %d pending microbatches)",
+                        e.grid.rank, e.grid.stage_id, self.timeout_s,
+                        e.schedule_position, e.global_steps, len(e._pending_dbg),
+                    )
+                time.sleep(5.0)
+            else:
+                time.sleep(min(remaining, 5.0))
+
+
 class PipelineEngine:
     def __init__(self, module: PipelineModule, config: TrainConfig, grid: ProcessGrid,
                  device: Optional[torch.device] = None):
@@ -105,6 +158,12 @@ class PipelineEngine:
         self.global_steps = 0
         self.skipped_steps = 0
         self._step_time = 0.0
+        # observability
+        self.timers = {k: 0.0 for k in ("forward", "backward", "p2p", "allreduce", "optimizer")}
+        self.schedule_position = "idle"
+        self._pending_dbg = ()
+        wd_s = float(os.environ.get("LPP_WATCHDOG_S", getattr(config, "watchdog_timeout_s", 0) or 0))
+        self.watchdog = _Watchdog(self, wd_s) if wd_s > 0 else None
 
     # ------------------------------------------------------------------
     @property
@@ -168,6 +227,8 @@ class PipelineEngine:
     def train_batch(self, data_iter: Iterator) -> torch.Tensor:
         """One optimizer step == ``micro_batches`` microbatches, 1F1B."""
         t0 = time.time()
+        if self.watchdog:
+            self.watchdog.arm()
         self.module.train()
         M = self.micro_batches
         P = self.grid.num_stages
@@ -179,17 +240,28 @@ class PipelineEngine:
         losses = []
 
         # ---- warmup forwards
-        for _ in range(warmup):
+        for wi in range(warmup):
+            self.schedule_position = f"warmup fwd {wi + 1}/{warmup}"
+            tp = time.time()
             x = self.p2p.recv_forward()
+            self.timers["p2p"] += time.time() - tp
+            tf = time.time()
             inp, handle, loss = self._forward_step(x, data_iter)
+            self.timers["forward"] += time.time() - tf
+            tp = time.time()
             self.p2p.send_forward(handle if not self.is_last_stage else None)
+            self.timers["p2p"] += time.time() - tp
             pending.append((inp, handle))
+            self._pending_dbg = tuple(range(len(pending)))
 
         x = self.p2p.recv_forward() if remaining > 0 else None
 
         # ---- steady 1F1B
         for i in range(remaining):
+            self.schedule_position = f"steady 1F1B {i + 1}/{remaining}"
+            tf = time.time()
             inp, handle, loss = self._forward_step(x, data_iter)
+            self.timers["forward"] += time.time() - tf
             pending.append((inp, handle))
             if loss is not None:
                 losses.append(loss)
@@ -198,7 +270,9 @@ class PipelineEngine:
             else:
                 recv_grad = self.p2p.send_forward_recv_backward(handle)
             b_inp, b_handle = pending.popleft()
+            tb = time.time()
             g = self._backward_step(b_inp, b_handle, recv_grad)
+            self.timers["backward"] += time.time() - tb
             last = i == remaining - 1
             if g is None:  # first stage sends nothing backward
                 x = None if last else self.p2p.recv_forward()
@@ -208,18 +282,27 @@ class PipelineEngine:
                 x = self.p2p.send_backward_recv_forward(g)
 
         # ---- cooldown backwards
-        for _ in range(warmup):
+        for ci in range(warmup):
+            self.schedule_position = f"cooldown bwd {ci + 1}/{warmup}"
             b_inp, b_handle = pending.popleft()
+            tp = time.time()
             recv_grad = self.p2p.recv_backward() if not self.is_last_stage else None
+            self.timers["p2p"] += time.time() - tp
+            tb = time.time()
             g = self._backward_step(b_inp, b_handle, recv_grad)
+            self.timers["backward"] += time.time() - tb
             if g is not None:
                 self.p2p.send_backward(g)
 
         # ---- boundary: DP all-reduce, clip, step
+        self.schedule_position = "optimizer"
         self._optimizer_step()
 
         loss_out = self._reduce_loss(losses)
         self._step_time = time.time() - t0
+        self.schedule_position = "idle"
+        if self.watchdog:
+            self.watchdog.disarm()
         return loss_out
 
     # ------------------------------------------------------------------
@@ -255,7 +338,10 @@ class PipelineEngine:
         flat.div_(self.grid.dp_degree)
 
     def _optimizer_step(self) -> None:
+        ta = time.time()
         self._allreduce_gradients()
+        self.timers["allreduce"] += time.time() - ta
+        t_o = time.time()
 
         inv_scale = 1.0
         if self.loss_scaler is not None:
@@ -283,11 +369,22 @@ class PipelineEngine:
         self.optimizer.zero_grad()
         self.lr_scheduler.step()
         self.global_steps += 1
+        self.timers["optimizer"] += time.time() - t_o
 
     # ------------------------------------------------------------------
     @property
     def last_step_time(self) -> float:
         return self._step_time
+
+    def timer_summary(self, reset: bool = True) -> dict:
+        """Host-side accumulated section times (seconds) since last reset.
+        Note these bracket ENQUEUE + sync points of async GPU work — use
+        rocprofv3 for device-side truth; these locate stalls per stage."""
+        out = dict(self.timers)
+        if reset:
+            for k in self.timers:
+                self.timers[k] = 0.0
+        return out
 
     def get_lr(self) -> float:
         return self.optimizer.lr

@@ -50,3 +50,35 @@ def test_fp16_loss_scaler_runs():
     it = sequential_loader(cfg)
     loss = engine.train_batch(it)
     assert torch.isfinite(torch.tensor(float(loss)))
+
+
+def test_watchdog_and_timers(monkeypatch):
+    """Watchdog arms/disarms around train_batch; timers accumulate."""
+    import lpp_amd.engine as eng
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.data import CausalLMCollator, SyntheticCausalLMDataset
+    from lpp_amd.models import get_layers_from_config, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+    import torch
+
+    monkeypatch.setenv("LPP_WATCHDOG_S", "300")
+    mcfg = model_config("tiny", num_layers=2, hidden_size=64, num_heads=4,
+                        intermediate_size=128, vocab_size=128, max_seq_len=32)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=32, dtype="fp32")
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    engine = eng.PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    assert engine.watchdog is not None
+    ds = SyntheticCausalLMDataset(8, 32, mcfg.vocab_size)
+    loader = torch.utils.data.DataLoader(ds, batch_size=2,
+                                         collate_fn=CausalLMCollator(32))
+    loss = engine.train_batch(iter(loader))
+    assert torch.isfinite(loss)
+    assert engine.schedule_position == "idle"
+    t = engine.timer_summary()
+    assert t["forward"] > 0 and t["backward"] > 0 and t["optimizer"] >= 0
+    # reset happened
+    assert engine.timers["forward"] == 0.0
