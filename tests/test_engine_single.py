@@ -63,7 +63,7 @@ def test_watchdog_and_timers(monkeypatch):
     import torch
 
     monkeypatch.setenv("LPP_WATCHDOG_S", "300")
-    mcfg = model_config("tiny", num_layers=2, hidden_size=64, num_heads=4,
+    mcfg = model_config("llama-tiny", num_layers=2, hidden_size=64, num_heads=4,
                         intermediate_size=128, vocab_size=128, max_seq_len=32)
     cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
                       gradient_accumulation_steps=2, seq_len=32, dtype="fp32")
