@@ -150,6 +150,8 @@ class TrainConfig:
     # Comm
     backend: str = "nccl"  # RCCL on ROCm; "gloo" for CPU tests
     p2p_overlap: bool = True
+    zero_stage: int = 0  # 1 = DP-sharded optimizer states (conf/...yaml:152-159)
+    watchdog_timeout_s: float = 0.0  # >0 arms the deadlock watchdog
     allreduce_bucket_mb: int = 200
 
     @property

@@ -143,12 +143,16 @@ class PipelineEngine:
         self.p2p = PipeP2P(grid, act_shape, comm_dtype, device)
 
         opt_cfg = config.optimizer
+        self.zero1 = bool(getattr(config, "zero_stage", 0) == 1 and grid.dp_degree > 1)
         self.optimizer = MixedPrecisionAdamW(
             self.module.parameters(),
             lr=opt_cfg.lr,
             betas=tuple(opt_cfg.betas),
             eps=opt_cfg.eps,
             weight_decay=opt_cfg.weight_decay,
+            shard_group=grid.dp_group if self.zero1 else None,
+            shard_rank=grid.dp_id if self.zero1 else 0,
+            shard_world=grid.dp_degree if self.zero1 else 1,
         )
         warmup = opt_cfg.warmup_steps or max(1, int(opt_cfg.warmup_proportion * opt_cfg.total_num_steps))
         self.lr_scheduler = WarmupDecayLR(
@@ -325,6 +329,18 @@ class PipelineEngine:
         if self.grid.dp_degree <= 1 or not dist.is_initialized():
             return
         flat = self.optimizer.flat_grads
+        if self.zero1:
+            # ZeRO-1: each DP rank only needs ITS shard of the summed grads.
+            opt = self.optimizer
+            if dist.get_backend(self.grid.dp_group) == "nccl":
+                shard = torch.empty_like(opt.grad_shard)
+                dist.reduce_scatter_tensor(shard, flat, op=dist.ReduceOp.SUM,
+                                           group=self.grid.dp_group)
+                opt.grad_shard.copy_(shard)
+            else:  # gloo: no reduce_scatter; all-reduce then slice locally
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=self.grid.dp_group)
+            opt.grad_shard.div_(self.grid.dp_degree)
+            return
         bucket_elems = max(1, self.config.allreduce_bucket_mb * 1024 * 1024 // 4)
         handles = []
         for off in range(0, flat.numel(), bucket_elems):
@@ -348,8 +364,12 @@ class PipelineEngine:
             inv_scale = 1.0 / self.loss_scaler.scale
 
         sq = self.optimizer.grad_sq_sum()
-        if self.grid.num_stages > 1 and dist.is_initialized():
-            dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.grid.pipe_group)
+        if dist.is_initialized():
+            if self.zero1:
+                # shards tile (dp x stage): sum across the whole world
+                dist.all_reduce(sq, op=dist.ReduceOp.SUM)
+            elif self.grid.num_stages > 1:
+                dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.grid.pipe_group)
         global_norm = (sq.float().sqrt() * inv_scale).item()
 
         if self.loss_scaler is not None:

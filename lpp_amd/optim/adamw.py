@@ -26,11 +26,21 @@ import math
 from typing import Iterable, List, Optional
 
 import torch
+import torch.distributed as dist
 
 from .. import ops
 
 
 class MixedPrecisionAdamW:
+    """``shard_world > 1`` enables ZeRO-1 semantics (optimizer-state
+    sharding across the DP group, conf/...yaml:152-159 / SURVEY.md §2.3):
+    parameters are flattened into one contiguous buffer, each DP rank owns
+    a 1/dp shard of the fp32 master + Adam moments, steps only its shard
+    against the (reduce-scattered) gradient shard, and all-gathers the
+    updated flat parameters.  Memory for optimizer state drops from
+    12 B/param to 12/dp B/param per rank.  Checkpoints of the optimizer
+    state are shard-local: resume requires the same dp_degree."""
+
     def __init__(
         self,
         params: Iterable[torch.nn.Parameter],
@@ -38,6 +48,9 @@ class MixedPrecisionAdamW:
         betas=(0.9, 0.99),
         eps: float = 1e-6,
         weight_decay: float = 0.001,
+        shard_group=None,
+        shard_rank: int = 0,
+        shard_world: int = 1,
     ):
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
@@ -47,25 +60,54 @@ class MixedPrecisionAdamW:
         self.eps = eps
         self.weight_decay = weight_decay
         self.step_count = 0
+        self.shard_group = shard_group
+        self.shard_rank = shard_rank
+        self.shard_world = shard_world
 
         dev = self.params[0].device
         total = sum(p.numel() for p in self.params)
+        pad = (-total) % max(shard_world, 1)
+        self.padded_total = total + pad
         # one flat fp32 grad buffer; per-param views exposed as p.main_grad
-        self.flat_grads = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.flat_grads = torch.zeros(self.padded_total, dtype=torch.float32, device=dev)
         self.masters: List[torch.Tensor] = []
         self.exp_avg: List[torch.Tensor] = []
         self.exp_avg_sq: List[torch.Tensor] = []
         self.grad_views: List[torch.Tensor] = []
-        off = 0
-        for p in self.params:
-            n = p.numel()
-            g = self.flat_grads[off : off + n].view_as(p)
-            p.main_grad = g
-            self.grad_views.append(g)
-            self.masters.append(p.detach().to(torch.float32).clone())
-            self.exp_avg.append(torch.zeros_like(self.masters[-1]))
-            self.exp_avg_sq.append(torch.zeros_like(self.masters[-1]))
-            off += n
+        self.flat_params: Optional[torch.Tensor] = None
+
+        if shard_world > 1:
+            # ZeRO-1: flatten params; shard-local fp32 state
+            pdtype = self.params[0].dtype
+            self.flat_params = torch.zeros(self.padded_total, dtype=pdtype, device=dev)
+            off = 0
+            for p in self.params:
+                n = p.numel()
+                self.flat_params[off : off + n].copy_(p.data.reshape(-1))
+                p.data = self.flat_params[off : off + n].view_as(p)
+                g = self.flat_grads[off : off + n].view_as(p)
+                p.main_grad = g
+                self.grad_views.append(g)
+                off += n
+            self.shard_size = self.padded_total // shard_world
+            lo = shard_rank * self.shard_size
+            self.shard_slice = slice(lo, lo + self.shard_size)
+            self.param_shard = self.flat_params[self.shard_slice]
+            self.masters = [self.param_shard.to(torch.float32).clone()]
+            self.exp_avg = [torch.zeros_like(self.masters[0])]
+            self.exp_avg_sq = [torch.zeros_like(self.masters[0])]
+            self.grad_shard = self.flat_grads[self.shard_slice]
+        else:
+            off = 0
+            for p in self.params:
+                n = p.numel()
+                g = self.flat_grads[off : off + n].view_as(p)
+                p.main_grad = g
+                self.grad_views.append(g)
+                self.masters.append(p.detach().to(torch.float32).clone())
+                self.exp_avg.append(torch.zeros_like(self.masters[-1]))
+                self.exp_avg_sq.append(torch.zeros_like(self.masters[-1]))
+                off += n
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._accumulate_hook) for p in self.params
         ]
@@ -77,9 +119,17 @@ class MixedPrecisionAdamW:
             p.grad = None
 
     # ------------------------------------------------------------------
+    @property
+    def is_sharded(self) -> bool:
+        return self.shard_world > 1
+
     def grad_sq_sum(self) -> torch.Tensor:
-        """Local sum of squared gradients (fp32 scalar tensor) — callers
-        all-reduce it across the pipe group for the global norm."""
+        """Local sum of squared gradients (fp32 scalar tensor).  Unsharded:
+        the full flat buffer (callers all-reduce across the PIPE group).
+        Sharded (ZeRO-1): this rank's gradient shard only (callers
+        all-reduce across the WORLD: shards tile the dp x stage space)."""
+        if self.is_sharded:
+            return (self.grad_shard * self.grad_shard).sum()
         return (self.flat_grads * self.flat_grads).sum()
 
     def zero_grad(self) -> None:
@@ -95,12 +145,19 @@ class MixedPrecisionAdamW:
         bias1 = 1.0 - beta1**t
         bias2 = 1.0 - beta2**t
 
+        if self.is_sharded:
+            plist = [self.param_shard]
+            glist = [self.grad_shard]
+        else:
+            plist = [p.data for p in self.params]
+            glist = self.grad_views
+
         if self.params[0].is_cuda and not ops.force_eager():
             ext = ops.extension()
             ext.fused_adamw(
-                [p.data for p in self.params],
+                plist,
                 self.masters,
-                self.grad_views,
+                glist,
                 self.exp_avg,
                 self.exp_avg_sq,
                 self.lr,
@@ -112,9 +169,10 @@ class MixedPrecisionAdamW:
                 bias2,
                 grad_scale,
             )
+            self._maybe_allgather_params()
             return
 
-        grads = self.grad_views
+        grads = glist
         if grad_scale != 1.0:
             torch._foreach_mul_(grads, grad_scale)
         # decoupled weight decay on master weights
@@ -129,12 +187,25 @@ class MixedPrecisionAdamW:
         torch._foreach_div_(denom, math.sqrt(bias2))
         torch._foreach_add_(denom, self.eps)
         torch._foreach_addcdiv_(self.masters, self.exp_avg, denom, value=-step_size)
-        for p, m in zip(self.params, self.masters):
-            p.data.copy_(m)
+        if self.is_sharded:
+            self.param_shard.copy_(self.masters[0])
+        else:
+            for p, m in zip(self.params, self.masters):
+                p.data.copy_(m)
+        self._maybe_allgather_params()
+
+    def _maybe_allgather_params(self) -> None:
+        if not self.is_sharded or not dist.is_initialized():
+            return
+        shards = [self.flat_params[i * self.shard_size : (i + 1) * self.shard_size]
+                  for i in range(self.shard_world)]
+        dist.all_gather(shards, self.param_shard.contiguous(), group=self.shard_group)
 
     # -- checkpoint state ---------------------------------------------------
     def state_dict(self) -> dict:
         return {
+            "shard_world": self.shard_world,
+            "shard_rank": self.shard_rank,
             "step_count": self.step_count,
             "lr": self.lr,
             "betas": self.betas,
@@ -146,6 +217,11 @@ class MixedPrecisionAdamW:
         }
 
     def load_state_dict(self, sd: dict) -> None:
+        if sd.get("shard_world", 1) != self.shard_world:
+            raise ValueError(
+                f"optimizer checkpoint sharded over {sd.get('shard_world', 1)} ranks, "
+                f"engine configured for {self.shard_world} (ZeRO-1 resume requires "
+                "the same dp_degree)")
         self.step_count = sd["step_count"]
         self.lr = sd.get("lr", self.lr)
         for dst, src in zip(self.masters, sd["masters"]):
@@ -156,5 +232,9 @@ class MixedPrecisionAdamW:
             dst.copy_(src.to(dst.device))
         # re-sync model params from masters
         with torch.no_grad():
-            for p, m in zip(self.params, self.masters):
-                p.data.copy_(m)
+            if self.is_sharded:
+                self.param_shard.copy_(self.masters[0])
+                self._maybe_allgather_params()
+            else:
+                for p, m in zip(self.params, self.masters):
+                    p.data.copy_(m)
