@@ -317,13 +317,36 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// dK/dV: 4 waves x 32 kv rows (lane&31 = own kv row), 256 threads,
-// 1 wave/SIMD (full 512-VGPR budget), q tiles of 32,
-// double-buffered Q (normal), dO (normal), Q^T, dO^T (stride-40 images),
-// L/Delta staged per tile. GQA: accumulate over the group's query heads.
-constexpr int KV_KW = 32, KV_WAVES = 4, KV_WG = 128, KV_QT = 32;
+// dK/dV: 8 waves x 16 kv rows (lane&15 = own kv row; 16x16x32 MFMA
+// fragments), 512 threads, 2 waves/SIMD with no spills.  Grid over
+// (S/128 kv blocks) x (B*HKV).  Per 32-row q tile (double-buffered):
+//   S    = mfma16(Q_lds, K_reg)   C[q regs][kv lane]
+//   dP   = mfma16(dO_lds, V_reg)  same layout
+//   P    = exp2(c*S - L[q]);  dS = P*(dP - delta[q])   (L/D staged)
+//   pack P, dS through a per-wave LDS tile into A[m=kv][k=q] fragments
+//   dV  += mfma16(pack(P),  dO^T image);  dK += mfma16(pack(dS), Q^T image)
+// GQA: the G query heads sharing a kv head accumulate in-register.
+constexpr int KV_KW = 16, KV_WG = 128, KV_QT = 32;
+constexpr int TS40 = 40;  // transposed-image row stride (odd word count)
 
-__global__ __launch_bounds__(256, 2) void attn_bwd_dkdv_kernel(
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// write the quad-transposed columns into a [128][TS40] image
+__device__ __forceinline__ void write_transposed40(char* img, int r0, int d0,
+                                                   const int dw[4]) {
+#pragma unroll
+  for (int e = 0; e < 2; ++e) {
+    const int d = d0 + e;
+    unsigned w01 = e ? (((unsigned)dw[0] >> 16) | ((unsigned)dw[1] & 0xffff0000u))
+                     : (((unsigned)dw[0] & 0xffffu) | ((unsigned)dw[1] << 16));
+    unsigned w23 = e ? (((unsigned)dw[2] >> 16) | ((unsigned)dw[3] & 0xffff0000u))
+                     : (((unsigned)dw[2] & 0xffffu) | ((unsigned)dw[3] << 16));
+    int2v pair = {(int)w01, (int)w23};
+    *reinterpret_cast<int2v*>(img + (d * TS40 + r0) * 2) = pair;
+  }
+}
+
+__global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
     const float* __restrict__ LSE2, const float* __restrict__ Delta,
@@ -331,19 +354,20 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkdv_kernel(
     int B, int S, int H, int HKV, float c, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // Q0@0 Q1@8K dO0@16K dO1@24K Qt0@32768 Qt1@+10240 dOt0@53248 dOt1@+10240
-  // L@73728 (2 x 32 f32) D@74240
+  // L@73728 D@73984 (2x32 f32 each) p_wave@74240 (8 x [16][40] bf16)
   auto q_lds = [&](int buf) -> char* { return smem + buf * 8192; };
   auto do_lds = [&](int buf) -> char* { return smem + 16384 + buf * 8192; };
-  auto qt_lds = [&](int buf) -> char* { return smem + 32768 + buf * 8192; };
-  auto dot_lds = [&](int buf) -> char* { return smem + 49152 + buf * 8192; };
-  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 65536) + buf * 32; };
-  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 65792) + buf * 32; };
+  auto qt_lds = [&](int buf) -> char* { return smem + 32768 + buf * 10240; };
+  auto dot_lds = [&](int buf) -> char* { return smem + 53248 + buf * 10240; };
+  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 73728) + buf * 32; };
+  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 73984) + buf * 32; };
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int lq = lane & 31;
-  const int hi2 = lane >> 5;
+  const int l16 = lane & 15;
+  const int hi4 = (lane >> 4) & 3;
+  short* pw = reinterpret_cast<short*>(smem + 74240) + wid * (16 * TS40);
 
   const int b = blockIdx.y / HKV;
   const int hkv = blockIdx.y % HKV;
@@ -352,138 +376,148 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkdv_kernel(
   const int64_t sHkvD = (int64_t)HKV * AB_D;
   const int64_t kv_base = (((int64_t)b * S) * HKV + hkv) * AB_D;
 
-  const int kvw0 = blockIdx.x * KV_WG + wid * KV_KW;  // wave's kv rows
-  const int kvrow = kvw0 + lq;                        // lane's kv row
+  const int kvw0 = blockIdx.x * KV_WG + wid * KV_KW;  // wave's 16 kv rows
+  const int kvrow = kvw0 + l16;                       // lane's kv row
 
-  // K/V B-fragments for this wave's 32 kv rows (lane n = kv row)
-  bf16x8 kf[8], vf[8];
+  // K/V B-fragments (16x16x32: lane n = l16 = kv row, k = hi4*8+j)
+  bf16x8 kf[4], vf[4];
   {
     const int64_t rb = kv_base + (int64_t)min(kvrow, S - 1) * sHkvD;
 #pragma unroll
-    for (int dc = 0; dc < 8; ++dc) {
-      kf[dc] = *reinterpret_cast<const bf16x8*>(&K[rb + dc * 16 + hi2 * 8]);
-      vf[dc] = *reinterpret_cast<const bf16x8*>(&V[rb + dc * 16 + hi2 * 8]);
+    for (int dc = 0; dc < 4; ++dc) {
+      kf[dc] = *reinterpret_cast<const bf16x8*>(&K[rb + dc * 32 + hi4 * 8]);
+      vf[dc] = *reinterpret_cast<const bf16x8*>(&V[rb + dc * 32 + hi4 * 8]);
     }
   }
 
-  f32x16 dv_acc[4], dk_acc[4];
+  f32x4 dv_acc[8], dk_acc[8];
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
+  for (int dt = 0; dt < 8; ++dt)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) { dv_acc[dt][r] = 0.f; dk_acc[dt][r] = 0.f; }
+    for (int r = 0; r < 4; ++r) { dv_acc[dt][r] = 0.f; dk_acc[dt][r] = 0.f; }
 
-  const int st_r = lane >> 4;
+  const int st_r = (lane >> 4) & 3;
   const int st_c = 8 * (lane & 15);
-  const int qstart = blockIdx.x * KV_WG;  // first q tile with any work
+  const int qstart = blockIdx.x * KV_WG;
 
   for (int g = 0; g < G; ++g) {
     const int h = hkv * G + g;
     const int64_t q_base = (((int64_t)b * S) * H + h) * AB_D;
     const int64_t ld_base = ((int64_t)b * H + h) * S;
 
-    auto ld_tile = [&](int qt0, int pass, bf16x8& qreg, bf16x8& dreg) {
-      const int row = min(qt0 + 4 * wid + 16 * pass + st_r, S - 1);
+    // one pass: 512 threads x 8 elems = one [32][128] tile
+    auto ld_tile = [&](int qt0, bf16x8& qreg, bf16x8& dreg) {
+      const int row = min(qt0 + 4 * wid + st_r, S - 1);
       const int64_t rb = q_base + (int64_t)row * sHD + st_c;
       qreg = *reinterpret_cast<const bf16x8*>(&Q[rb]);
       dreg = *reinterpret_cast<const bf16x8*>(&dO[rb]);
     };
-    auto write_tile = [&](int buf, int qt0, int pass, bf16x8 qreg, bf16x8 dreg) {
-      const int r = 4 * wid + 16 * pass + st_r;
+    auto write_tile = [&](int buf, int qt0, bf16x8 qreg, bf16x8 dreg) {
+      const int r = 4 * wid + st_r;
       *reinterpret_cast<bf16x8*>(q_lds(buf) + bswz(r, st_c)) = qreg;
       *reinterpret_cast<bf16x8*>(do_lds(buf) + bswz(r, st_c)) = dreg;
       int dw[4];
       quad_transpose(qreg, st_r, dw);
-      write_transposed32(qt_lds(buf), 4 * wid + 16 * pass, st_c + 2 * st_r, dw);
+      write_transposed40(qt_lds(buf), 4 * wid, st_c + 2 * st_r, dw);
       quad_transpose(dreg, st_r, dw);
-      write_transposed32(dot_lds(buf), 4 * wid + 16 * pass, st_c + 2 * st_r, dw);
-      if (pass == 0 && tid < 64) {  // L/Delta for the tile's 32 q rows
+      write_transposed40(dot_lds(buf), 4 * wid, st_c + 2 * st_r, dw);
+      if (tid < 64) {
         const int qi = min(qt0 + (tid & 31), S - 1);
         if (tid < 32) l_buf(buf)[tid] = LSE2[ld_base + qi];
         else d_buf(buf)[tid & 31] = Delta[ld_base + qi];
       }
     };
 
-    bf16x8 qreg0, qreg1, dreg0, dreg1;
-    ld_tile(qstart, 0, qreg0, dreg0);
-    ld_tile(qstart, 1, qreg1, dreg1);
-    write_tile(0, qstart, 0, qreg0, dreg0);
-    write_tile(0, qstart, 1, qreg1, dreg1);
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+      __builtin_amdgcn_s_setprio(1);
+    bf16x8 qreg, dreg;
+    ld_tile(qstart, qreg, dreg);
+    write_tile(0, qstart, qreg, dreg);
     __syncthreads();
 
     for (int qt0 = qstart, cur = 0; qt0 < S; qt0 += KV_QT, cur ^= 1) {
       const bool have_next = qt0 + KV_QT < S;
-      if (have_next) {
-        ld_tile(qt0 + KV_QT, 0, qreg0, dreg0);
-        ld_tile(qt0 + KV_QT, 1, qreg1, dreg1);
-      }
+      if (have_next) ld_tile(qt0 + KV_QT, qreg, dreg);
 
-      if (qt0 + KV_QT - 1 >= kvw0) {  // causal: some q in tile reaches this wave
-        f32x16 st, dpt;
+      if (qt0 + KV_QT - 1 >= kvw0) {
+        f32x4 st[2], dpt[2];
 #pragma unroll
-        for (int r = 0; r < 16; ++r) { st[r] = 0.f; dpt[r] = 0.f; }
+        for (int qs = 0; qs < 2; ++qs)
 #pragma unroll
-        for (int dc = 0; dc < 8; ++dc) {
-          const bf16x8 aq = *reinterpret_cast<const bf16x8*>(
-              q_lds(cur) + bswz(lq, dc * 16 + hi2 * 8));
-          const bf16x8 ad = *reinterpret_cast<const bf16x8*>(
-              do_lds(cur) + bswz(lq, dc * 16 + hi2 * 8));
-          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[dc], st, 0, 0, 0);
-          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[dc], dpt, 0, 0, 0);
+          for (int r = 0; r < 4; ++r) { st[qs][r] = 0.f; dpt[qs][r] = 0.f; }
+#pragma unroll
+        for (int dc = 0; dc < 4; ++dc) {
+#pragma unroll
+          for (int qs = 0; qs < 2; ++qs) {
+            const bf16x8 aq = *reinterpret_cast<const bf16x8*>(
+                q_lds(cur) + bswz(qs * 16 + l16, dc * 32 + hi4 * 8));
+            const bf16x8 ad = *reinterpret_cast<const bf16x8*>(
+                do_lds(cur) + bswz(qs * 16 + l16, dc * 32 + hi4 * 8));
+            st[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, kf[dc], st[qs], 0, 0, 0);
+            dpt[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ad, vf[dc], dpt[qs], 0, 0, 0);
+          }
         }
 
         const bool need_mask = (qt0 < kvw0 + KV_KW) || (qt0 + KV_QT > S);
-        float p[16], ds[16];
+        float p[2][4], ds[2][4];
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi2;  // q row of reg r
-          const float Lr = l_buf(cur)[qr];
-          const float Dr = d_buf(cur)[qr];
-          bool ok = (kvrow < S);
-          if (need_mask) {
-            const int q = qt0 + qr;
-            ok = ok && (q >= kvrow) && (q < S);
+        for (int qs = 0; qs < 2; ++qs)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int qr = qs * 16 + hi4 * 4 + r;  // q row within the tile
+            const float Lr = l_buf(cur)[qr];
+            const float Dr = d_buf(cur)[qr];
+            bool ok = (kvrow < S);
+            if (need_mask) {
+              const int q = qt0 + qr;
+              ok = ok && (q >= kvrow) && (q < S);
+            }
+            p[qs][r] = ok ? exp2f(fmaf(st[qs][r], c, -Lr)) : 0.f;
+            ds[qs][r] = p[qs][r] * (dpt[qs][r] - Dr);
           }
-          p[r] = ok ? exp2f(fmaf(st[r], c, -Lr)) : 0.f;
-          ds[r] = p[r] * (dpt[r] - Dr);
-        }
 
-        bf16x8 pa[2], dsa[2];
-        pack_pair(p, pa);
-        pack_pair(ds, dsa);
+        // route P and dS through the per-wave LDS tile -> A-fragments
+        bf16x8 pa, dsa;
+#pragma unroll
+        for (int qs = 0; qs < 2; ++qs)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            pw[l16 * TS40 + qs * 16 + hi4 * 4 + r] = f2bf_(p[qs][r]);
+        pa = *reinterpret_cast<const bf16x8*>(&pw[l16 * TS40 + hi4 * 8]);
+#pragma unroll
+        for (int qs = 0; qs < 2; ++qs)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            pw[l16 * TS40 + qs * 16 + hi4 * 4 + r] = f2bf_(ds[qs][r]);
+        dsa = *reinterpret_cast<const bf16x8*>(&pw[l16 * TS40 + hi4 * 8]);
 
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
-          const int d = dt * 32 + lq;
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks) {
-            const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
-                dot_lds(cur) + tswz32(d, ks * 16 + hi2 * 8));
-            const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-                qt_lds(cur) + tswz32(d, ks * 16 + hi2 * 8));
-            dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], bd, dv_acc[dt], 0, 0, 0);
-            dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[ks], bq, dk_acc[dt], 0, 0, 0);
-          }
+        for (int dt = 0; dt < 8; ++dt) {
+          const int d = dt * 16 + l16;
+          const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
+              dot_lds(cur) + (d * TS40 + hi4 * 8) * 2);
+          const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+              qt_lds(cur) + (d * TS40 + hi4 * 8) * 2);
+          dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bd, dv_acc[dt], 0, 0, 0);
+          dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bq, dk_acc[dt], 0, 0, 0);
         }
       }
 
-      if (have_next) {
-        write_tile(cur ^ 1, qt0 + KV_QT, 0, qreg0, dreg0);
-        write_tile(cur ^ 1, qt0 + KV_QT, 1, qreg1, dreg1);
-      }
+      if (have_next) write_tile(cur ^ 1, qt0 + KV_QT, qreg, dreg);
       __syncthreads();
     }
     __syncthreads();  // buffer 0 reuse across g
   }
 
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = kvw0 + (r & 3) + 8 * (r >> 2) + 4 * hi2;
+  for (int r = 0; r < 4; ++r) {
+    const int row = kvw0 + hi4 * 4 + r;
     if (row >= S) continue;
     const int64_t rb = kv_base + (int64_t)row * sHkvD;
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      dV[rb + dt * 32 + lq] = f2bf_(dv_acc[dt][r]);
-      dK[rb + dt * 32 + lq] = f2bf_(dk_acc[dt][r] * scale);
+    for (int dt = 0; dt < 8; ++dt) {
+      dV[rb + dt * 16 + l16] = f2bf_(dv_acc[dt][r]);
+      dK[rb + dt * 16 + l16] = f2bf_(dk_acc[dt][r] * scale);
     }
   }
 }
@@ -530,8 +564,8 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   }
   {
     const int kvblocks = (S + lpp::KV_WG - 1) / lpp::KV_WG;
-    const size_t lds = 66048;
-    hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(256), lds,
+    const size_t lds = 84480;
+    hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(512), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
                        lse2.data_ptr<float>(), delta.data_ptr<float>(),

@@ -163,6 +163,11 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[dt][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;
 
+  // static wave priority for the younger half (guide T5 static form):
+  // must be wave-uniform via readfirstlane or s_setprio applies to all waves
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
   // ---- prologue: stage tile 0 ----
   bf16x8 kreg0, kreg1, vreg0, vreg1;
   ld_tile(0, 0, kreg0, vreg0);
