@@ -77,6 +77,30 @@ MODEL_PRESETS = {
     "llama-65b": _preset(
         "llama-65b", hidden_size=8192, intermediate_size=22016, num_layers=80, num_heads=64
     ),
+    # LLaMA-2 family (GQA on 70B; rope theta 1e4, 4k ctx)
+    "llama2-7b": _preset(
+        "llama2-7b", hidden_size=4096, intermediate_size=11008, num_layers=32,
+        num_heads=32, max_seq_len=4096
+    ),
+    "llama2-13b": _preset(
+        "llama2-13b", hidden_size=5120, intermediate_size=13824, num_layers=40,
+        num_heads=40, max_seq_len=4096
+    ),
+    "llama2-70b": _preset(
+        "llama2-70b", hidden_size=8192, intermediate_size=28672, num_layers=80,
+        num_heads=64, num_kv_heads=8, max_seq_len=4096
+    ),
+    # LLaMA-3 family (GQA, 128k vocab, rope theta 5e5)
+    "llama3-8b": _preset(
+        "llama3-8b", hidden_size=4096, intermediate_size=14336, num_layers=32,
+        num_heads=32, num_kv_heads=8, vocab_size=128256, rope_theta=500000.0,
+        max_seq_len=8192
+    ),
+    "llama3-70b": _preset(
+        "llama3-70b", hidden_size=8192, intermediate_size=28672, num_layers=80,
+        num_heads=64, num_kv_heads=8, vocab_size=128256, rope_theta=500000.0,
+        max_seq_len=8192
+    ),
     # Tiny model for CPU tests.
     "llama-tiny": _preset(
         "llama-tiny",

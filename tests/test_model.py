@@ -51,3 +51,20 @@ def test_activation_checkpointing_same_output():
     l1.backward(); l2.backward()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1.grad, p2.grad, atol=1e-6)
+
+
+def test_gqa_presets():
+    """LLaMA-2/3 GQA presets build and the kv projections are narrower."""
+    from lpp_amd.config import model_config
+    from lpp_amd.models import LlamaForCausalLM, loss_fn
+    import torch
+
+    mcfg = model_config("llama2-70b", num_layers=1, hidden_size=128, num_heads=8,
+                        num_kv_heads=2, intermediate_size=256, vocab_size=128,
+                        max_seq_len=64)
+    model = LlamaForCausalLM(mcfg)
+    attn = model.layers[1].self_attn
+    assert attn.k_proj.weight.shape[0] == 2 * (128 // 8)
+    ids = torch.randint(0, 128, (2, 32))
+    loss = model.compute_loss(ids, ids.clone())
+    assert torch.isfinite(loss)
