@@ -326,24 +326,56 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
 //   pack P, dS through a per-wave LDS tile into A[m=kv][k=q] fragments
 //   dV  += mfma16(pack(P),  dO^T image);  dK += mfma16(pack(dS), Q^T image)
 // GQA: the G query heads sharing a kv head accumulate in-register.
-constexpr int KV_KW = 16, KV_WG = 128, KV_QT = 32;
+constexpr int KV_KW = 16, KV_WG = 128, KV_QT = 64;
+constexpr int TS72 = 72;  // 64-row transposed-image stride (odd word count)
 constexpr int TS40 = 40;  // transposed-image row stride (odd word count)
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-// write the quad-transposed columns into a [128][TS40] image
-__device__ __forceinline__ void write_transposed40(char* img, int r0, int d0,
+// write the quad-transposed columns into a [128][TS72] image.
+// Column order is staggered by lane ((d0>>4)+jj)&1 so each wave store
+// instruction covers both d-parities: 4-way banked instead of 8-way.
+__device__ __forceinline__ void write_transposed72(char* img, int r0, int d0,
                                                    const int dw[4]) {
 #pragma unroll
-  for (int e = 0; e < 2; ++e) {
+  for (int jj = 0; jj < 2; ++jj) {
+    const int e = ((d0 >> 4) + jj) & 1;
     const int d = d0 + e;
     unsigned w01 = e ? (((unsigned)dw[0] >> 16) | ((unsigned)dw[1] & 0xffff0000u))
                      : (((unsigned)dw[0] & 0xffffu) | ((unsigned)dw[1] << 16));
     unsigned w23 = e ? (((unsigned)dw[2] >> 16) | ((unsigned)dw[3] & 0xffff0000u))
                      : (((unsigned)dw[2] & 0xffffu) | ((unsigned)dw[3] << 16));
     int2v pair = {(int)w01, (int)w23};
-    *reinterpret_cast<int2v*>(img + (d * TS40 + r0) * 2) = pair;
+    *reinterpret_cast<int2v*>(img + (d * 72 + r0) * 2) = pair;
   }
+}
+
+
+// Pack a (lane = n-axis, regs = crow16 q-axis) C-layout value set into the
+// 16x16x32 MFMA A-fragment A[m = lane&15][k = (lane>>4)*8 + j] with pure
+// cross-lane VALU ops (no LDS):
+//   sources: X_qs = cvt_pk(p[qs][0], p[qs][1]), Y_qs = cvt_pk(p[qs][2], p[qs][3])
+//   a = permlane32_swap(X0, X1); c  = permlane16_swap(a[0], a[0]);
+//   c' = permlane16_swap(a[1], a[1])
+//   d0 = (hi4&1) ? c'[0] : a[0];   d2 = (hi4&1) ? a[1] : c[1]
+// (and the same for the Y family giving d1, d3).
+__device__ __forceinline__ bf16x8 pack_frag16(const float p[2][4], int hi4_odd) {
+  unsigned X0 = cvt_pk_bf16_(p[0][0], p[0][1]);
+  unsigned Y0 = cvt_pk_bf16_(p[0][2], p[0][3]);
+  unsigned X1 = cvt_pk_bf16_(p[1][0], p[1][1]);
+  unsigned Y1 = cvt_pk_bf16_(p[1][2], p[1][3]);
+  const int2v ax = __builtin_amdgcn_permlane32_swap((int)X0, (int)X1, false, false);
+  const int2v cx = __builtin_amdgcn_permlane16_swap(ax[0], ax[0], false, false);
+  const int2v cpx = __builtin_amdgcn_permlane16_swap(ax[1], ax[1], false, false);
+  const int2v ay = __builtin_amdgcn_permlane32_swap((int)Y0, (int)Y1, false, false);
+  const int2v cy = __builtin_amdgcn_permlane16_swap(ay[0], ay[0], false, false);
+  const int2v cpy = __builtin_amdgcn_permlane16_swap(ay[1], ay[1], false, false);
+  int w[4];
+  w[0] = hi4_odd ? cpx[0] : ax[0];
+  w[1] = hi4_odd ? cpy[0] : ay[0];
+  w[2] = hi4_odd ? ax[1] : cx[1];
+  w[3] = hi4_odd ? ay[1] : cy[1];
+  return *reinterpret_cast<const bf16x8*>(w);
 }
 
 __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
@@ -353,21 +385,22 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     short* __restrict__ dK, short* __restrict__ dV,
     int B, int S, int H, int HKV, float c, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // Q0@0 Q1@8K dO0@16K dO1@24K Qt0@32768 Qt1@+10240 dOt0@53248 dOt1@+10240
-  // L@73728 D@73984 (2x32 f32 each) p_wave@74240 (8 x [16][40] bf16)
-  auto q_lds = [&](int buf) -> char* { return smem + buf * 8192; };
-  auto do_lds = [&](int buf) -> char* { return smem + 16384 + buf * 8192; };
-  auto qt_lds = [&](int buf) -> char* { return smem + 32768 + buf * 10240; };
-  auto dot_lds = [&](int buf) -> char* { return smem + 53248 + buf * 10240; };
-  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 73728) + buf * 32; };
-  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 73984) + buf * 32; };
+  // 64-row double-buffered tiles:
+  // Q0@0 Q1@16K dO0@32K dO1@48K Qt0@64K Qt1@+18432 dOt0@100864... all 16-B
+  // aligned: Qt/dOt are [128][72]*2B = 18432 B each.
+  auto q_lds = [&](int buf) -> char* { return smem + buf * 16384; };
+  auto do_lds = [&](int buf) -> char* { return smem + 32768 + buf * 16384; };
+  auto qt_lds = [&](int buf) -> char* { return smem + 65536 + buf * 18432; };
+  auto dot_lds = [&](int buf) -> char* { return smem + 102400 + buf * 18432; };
+  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 139264) + buf * 64; };
+  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 139776) + buf * 64; };
+
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
   const int l16 = lane & 15;
   const int hi4 = (lane >> 4) & 3;
-  short* pw = reinterpret_cast<short*>(smem + 74240) + wid * (16 * TS40);
 
   const int b = blockIdx.y / HKV;
   const int hkv = blockIdx.y % HKV;
@@ -405,105 +438,108 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     const int64_t q_base = (((int64_t)b * S) * H + h) * AB_D;
     const int64_t ld_base = ((int64_t)b * H + h) * S;
 
-    // one pass: 512 threads x 8 elems = one [32][128] tile
-    auto ld_tile = [&](int qt0, bf16x8& qreg, bf16x8& dreg) {
-      const int row = min(qt0 + 4 * wid + st_r, S - 1);
+    // two passes: 512 threads x 8 elems x 2 = one [64][128] tile
+    auto ld_tile = [&](int qt0, int pass, bf16x8& qreg, bf16x8& dreg) {
+      const int row = min(qt0 + 4 * wid + 32 * pass + st_r, S - 1);
       const int64_t rb = q_base + (int64_t)row * sHD + st_c;
       qreg = *reinterpret_cast<const bf16x8*>(&Q[rb]);
       dreg = *reinterpret_cast<const bf16x8*>(&dO[rb]);
     };
-    auto write_tile = [&](int buf, int qt0, bf16x8 qreg, bf16x8 dreg) {
-      const int r = 4 * wid + st_r;
+    auto write_tile = [&](int buf, int qt0, int pass, bf16x8 qreg, bf16x8 dreg) {
+      const int r = 4 * wid + 32 * pass + st_r;
       *reinterpret_cast<bf16x8*>(q_lds(buf) + bswz(r, st_c)) = qreg;
       *reinterpret_cast<bf16x8*>(do_lds(buf) + bswz(r, st_c)) = dreg;
       int dw[4];
       quad_transpose(qreg, st_r, dw);
-      write_transposed40(qt_lds(buf), 4 * wid, st_c + 2 * st_r, dw);
+      write_transposed72(qt_lds(buf), r - st_r, st_c + 2 * st_r, dw);
       quad_transpose(dreg, st_r, dw);
-      write_transposed40(dot_lds(buf), 4 * wid, st_c + 2 * st_r, dw);
-      if (tid < 64) {
-        const int qi = min(qt0 + (tid & 31), S - 1);
-        if (tid < 32) l_buf(buf)[tid] = LSE2[ld_base + qi];
-        else d_buf(buf)[tid & 31] = Delta[ld_base + qi];
+      write_transposed72(dot_lds(buf), r - st_r, st_c + 2 * st_r, dw);
+      if (pass == 0 && tid < 128) {
+        const int qi = min(qt0 + (tid & 63), S - 1);
+        if (tid < 64) l_buf(buf)[tid] = LSE2[ld_base + qi];
+        else d_buf(buf)[tid & 63] = Delta[ld_base + qi];
       }
     };
 
     if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
       __builtin_amdgcn_s_setprio(1);
-    bf16x8 qreg, dreg;
-    ld_tile(qstart, qreg, dreg);
-    write_tile(0, qstart, qreg, dreg);
+    bf16x8 qreg0, dreg0, qreg1, dreg1;
+    ld_tile(qstart, 0, qreg0, dreg0);
+    ld_tile(qstart, 1, qreg1, dreg1);
+    write_tile(0, qstart, 0, qreg0, dreg0);
+    write_tile(0, qstart, 1, qreg1, dreg1);
     __syncthreads();
 
     for (int qt0 = qstart, cur = 0; qt0 < S; qt0 += KV_QT, cur ^= 1) {
       const bool have_next = qt0 + KV_QT < S;
-      if (have_next) ld_tile(qt0 + KV_QT, qreg, dreg);
+      if (have_next) {
+        ld_tile(qt0 + KV_QT, 0, qreg0, dreg0);
+        ld_tile(qt0 + KV_QT, 1, qreg1, dreg1);
+      }
 
       if (qt0 + KV_QT - 1 >= kvw0) {
-        f32x4 st[2], dpt[2];
 #pragma unroll
-        for (int qs = 0; qs < 2; ++qs)
+        for (int half = 0; half < 2; ++half) {
+          const int qh0 = qt0 + half * 32;
+          if (qh0 + 31 < kvw0) continue;
+          f32x4 st[2], dpt[2];
 #pragma unroll
-          for (int r = 0; r < 4; ++r) { st[qs][r] = 0.f; dpt[qs][r] = 0.f; }
+          for (int qs = 0; qs < 2; ++qs)
 #pragma unroll
-        for (int dc = 0; dc < 4; ++dc) {
+            for (int r = 0; r < 4; ++r) { st[qs][r] = 0.f; dpt[qs][r] = 0.f; }
+#pragma unroll
+          for (int dc = 0; dc < 4; ++dc) {
+#pragma unroll
+            for (int qs = 0; qs < 2; ++qs) {
+              const bf16x8 aq = *reinterpret_cast<const bf16x8*>(
+                  q_lds(cur) + bswz(half * 32 + qs * 16 + l16, dc * 32 + hi4 * 8));
+              const bf16x8 ad = *reinterpret_cast<const bf16x8*>(
+                  do_lds(cur) + bswz(half * 32 + qs * 16 + l16, dc * 32 + hi4 * 8));
+              st[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, kf[dc], st[qs], 0, 0, 0);
+              dpt[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ad, vf[dc], dpt[qs], 0, 0, 0);
+            }
+          }
+
+          const bool need_mask = (qh0 < kvw0 + KV_KW) || (qh0 + 32 > S);
+          float p[2][4], ds[2][4];
 #pragma unroll
           for (int qs = 0; qs < 2; ++qs) {
-            const bf16x8 aq = *reinterpret_cast<const bf16x8*>(
-                q_lds(cur) + bswz(qs * 16 + l16, dc * 32 + hi4 * 8));
-            const bf16x8 ad = *reinterpret_cast<const bf16x8*>(
-                do_lds(cur) + bswz(qs * 16 + l16, dc * 32 + hi4 * 8));
-            st[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq, kf[dc], st[qs], 0, 0, 0);
-            dpt[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ad, vf[dc], dpt[qs], 0, 0, 0);
-          }
-        }
-
-        const bool need_mask = (qt0 < kvw0 + KV_KW) || (qt0 + KV_QT > S);
-        float p[2][4], ds[2][4];
+            const int q0 = half * 32 + qs * 16 + hi4 * 4;
+            const f32x4 Lr = *reinterpret_cast<const f32x4*>(&l_buf(cur)[q0]);
+            const f32x4 Dr = *reinterpret_cast<const f32x4*>(&d_buf(cur)[q0]);
 #pragma unroll
-        for (int qs = 0; qs < 2; ++qs)
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int qr = qs * 16 + hi4 * 4 + r;  // q row within the tile
-            const float Lr = l_buf(cur)[qr];
-            const float Dr = d_buf(cur)[qr];
-            bool ok = (kvrow < S);
-            if (need_mask) {
-              const int q = qt0 + qr;
-              ok = ok && (q >= kvrow) && (q < S);
+            for (int r = 0; r < 4; ++r) {
+              bool ok = (kvrow < S);
+              if (need_mask) {
+                const int q = qt0 + q0 + r;
+                ok = ok && (q >= kvrow) && (q < S);
+              }
+              p[qs][r] = ok ? exp2f(fmaf(st[qs][r], c, -Lr[r])) : 0.f;
+              ds[qs][r] = p[qs][r] * (dpt[qs][r] - Dr[r]);
             }
-            p[qs][r] = ok ? exp2f(fmaf(st[qs][r], c, -Lr)) : 0.f;
-            ds[qs][r] = p[qs][r] * (dpt[qs][r] - Dr);
           }
 
-        // route P and dS through the per-wave LDS tile -> A-fragments
-        bf16x8 pa, dsa;
-#pragma unroll
-        for (int qs = 0; qs < 2; ++qs)
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            pw[l16 * TS40 + qs * 16 + hi4 * 4 + r] = f2bf_(p[qs][r]);
-        pa = *reinterpret_cast<const bf16x8*>(&pw[l16 * TS40 + hi4 * 8]);
-#pragma unroll
-        for (int qs = 0; qs < 2; ++qs)
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            pw[l16 * TS40 + qs * 16 + hi4 * 4 + r] = f2bf_(ds[qs][r]);
-        dsa = *reinterpret_cast<const bf16x8*>(&pw[l16 * TS40 + hi4 * 8]);
+          // cross-lane permlane pack (no LDS traffic)
+          const bf16x8 pa = pack_frag16(p, hi4 & 1);
+          const bf16x8 dsa = pack_frag16(ds, hi4 & 1);
 
 #pragma unroll
-        for (int dt = 0; dt < 8; ++dt) {
-          const int d = dt * 16 + l16;
-          const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
-              dot_lds(cur) + (d * TS40 + hi4 * 8) * 2);
-          const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-              qt_lds(cur) + (d * TS40 + hi4 * 8) * 2);
-          dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bd, dv_acc[dt], 0, 0, 0);
-          dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bq, dk_acc[dt], 0, 0, 0);
+          for (int dt = 0; dt < 8; ++dt) {
+            const int d = dt * 16 + l16;
+            const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
+                dot_lds(cur) + (d * TS72 + half * 32 + hi4 * 8) * 2);
+            const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+                qt_lds(cur) + (d * TS72 + half * 32 + hi4 * 8) * 2);
+            dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bd, dv_acc[dt], 0, 0, 0);
+            dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bq, dk_acc[dt], 0, 0, 0);
+          }
         }
       }
 
-      if (have_next) write_tile(cur ^ 1, qt0 + KV_QT, qreg, dreg);
+      if (have_next) {
+        write_tile(cur ^ 1, qt0 + KV_QT, 0, qreg0, dreg0);
+        write_tile(cur ^ 1, qt0 + KV_QT, 1, qreg1, dreg1);
+      }
       __syncthreads();
     }
     __syncthreads();  // buffer 0 reuse across g
@@ -564,7 +600,7 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   }
   {
     const int kvblocks = (S + lpp::KV_WG - 1) / lpp::KV_WG;
-    const size_t lds = 84480;
+    const size_t lds = 140288;
     hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(512), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
