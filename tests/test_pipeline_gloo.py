@@ -48,3 +48,20 @@ def test_dp2_pure_data_parallel():
 def test_pp2_all_ranks_same_loss():
     got = run_dist(2, run_steps, 2, 2, 4)
     assert got[0] == pytest.approx(got[1], abs=1e-6)
+
+
+def test_hybrid_pp2_dp2_matches_single_process():
+    """world 4 = PP2 x DP2 (the reference's hybrid mode, README.md:39-42):
+    same example set per step as a single process with gas doubled, so the
+    loss trajectory must match within fp tolerance."""
+    base = _single_process_baseline(steps=3, gas=8)
+    got = run_dist(4, run_steps, 2, 3, 4)  # stages=2 -> dp=2, gas=4/rank
+    for r in range(4):
+        for a, b in zip(base, got[r]):
+            assert abs(a - b) < 1e-3, (base, got[r])
+
+
+def test_hybrid_all_ranks_agree():
+    got = run_dist(4, run_steps, 2, 2, 4)
+    for r in range(1, 4):
+        assert got[0] == pytest.approx(got[r], abs=1e-6)
