@@ -174,6 +174,9 @@ class TrainConfig:
     # Comm
     backend: str = "nccl"  # RCCL on ROCm; "gloo" for CPU tests
     p2p_overlap: bool = True
+    eval_steps: int = 0  # run a forward-only eval pass every N steps
+    eval_micro_batches: int = 8
+    checkpoint_layers_per_stage: int = -1  # -1 = activation_checkpoint_interval rules
     zero_stage: int = 0  # 1 = DP-sharded optimizer states (conf/...yaml:152-159)
     watchdog_timeout_s: float = 0.0  # >0 arms the deadlock watchdog
     allreduce_bucket_mb: int = 200

@@ -309,6 +309,31 @@ class PipelineEngine:
             self.watchdog.disarm()
         return loss_out
 
+    @torch.no_grad()
+    def eval_batch(self, data_iter: Iterator, micro_batches: Optional[int] = None) -> torch.Tensor:
+        """Forward-only pipeline pass over ``micro_batches`` microbatches;
+        returns the mean loss on every rank.  Fills the gap the reference
+        leaves open (its config names an evaluator class that does not
+        exist and train() never calls one — SURVEY.md §2.9)."""
+        self.module.eval()
+        M = micro_batches or self.micro_batches
+        losses = []
+        for _ in range(M):
+            x = self.p2p.recv_forward()
+            if self.is_first_stage:
+                batch = self._next_batch(data_iter)
+                x = batch["input_ids"]
+            out = self.module(x)
+            self.p2p.send_forward(out if not self.is_last_stage else None)
+            if self.is_last_stage:
+                if self.is_first_stage:
+                    labels = batch["labels"]
+                else:
+                    labels = self._next_batch(data_iter)["labels"]
+                losses.append(self.module.loss_fn(out, labels).detach())
+        self.module.train()
+        return self._reduce_loss(losses)
+
     # ------------------------------------------------------------------
     def _reduce_loss(self, losses) -> torch.Tensor:
         if self.is_last_stage:

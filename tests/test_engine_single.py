@@ -82,3 +82,31 @@ def test_watchdog_and_timers(monkeypatch):
     assert t["forward"] > 0 and t["backward"] > 0 and t["optimizer"] >= 0
     # reset happened
     assert engine.timers["forward"] == 0.0
+
+
+def test_eval_batch_forward_only():
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset
+    from lpp_amd.engine import PipelineEngine
+    from lpp_amd.models import get_layers_from_config, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+    import torch
+
+    mcfg = model_config("llama-tiny", num_layers=2, max_seq_len=32)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=32, dtype="fp32")
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    engine = PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    ds = SyntheticCausalLMDataset(16, 32, mcfg.vocab_size)
+    it = iter(RepeatingLoader(torch.utils.data.DataLoader(
+        ds, batch_size=2, collate_fn=CausalLMCollator(32))))
+    before = [p.clone() for p in module.parameters()]
+    ev = engine.eval_batch(it, micro_batches=3)
+    assert torch.isfinite(ev) and ev > 0
+    # eval must not touch parameters or leave grads
+    for p0, p1 in zip(before, module.parameters()):
+        assert torch.equal(p0, p1)
+    assert engine.module.training

@@ -65,3 +65,19 @@ def test_hybrid_all_ranks_agree():
     got = run_dist(4, run_steps, 2, 2, 4)
     for r in range(1, 4):
         assert got[0] == pytest.approx(got[r], abs=1e-6)
+
+
+def _eval_run(rank, world, steps=1):
+    from tests.engine_utils import make_config, build_engine, sequential_loader
+
+    cfg = make_config(num_stages=world, gas=4)
+    engine = build_engine(cfg, rank, world)
+    it = sequential_loader(cfg) if (engine.is_first_stage or engine.is_last_stage) else None
+    return float(engine.eval_batch(it, micro_batches=4))
+
+
+def test_eval_batch_pp2():
+    """Forward-only eval over a 2-stage pipeline: both ranks get the loss."""
+    got = run_dist(2, _eval_run)
+    assert got[0] == pytest.approx(got[1], abs=1e-6)
+    assert got[0] > 0
