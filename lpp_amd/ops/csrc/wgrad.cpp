@@ -92,10 +92,10 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
       size_t ws = kWorkspaceBytes;
       LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceSetAttribute(
           pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-      hipblasLtMatmulHeuristicResult_t results[16];
+      hipblasLtMatmulHeuristicResult_t results[48];
       int found = 0;
       LPP_CHECK_BLASLT(hipblasLtMatmulAlgoGetHeuristic(
-          lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 16, results,
+          lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 48, results,
           &found));
       hipblasLtMatmulPreferenceDestroy(pref);
       TORCH_CHECK(found > 0, "hipblaslt: no algo for wgrad shape [", out, ",", in,
