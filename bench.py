@@ -98,7 +98,8 @@ def main() -> int:
         # shrink the microbatch count (pipeline bubble) at fixed tokens/step
         args.micro_batch_size = {1: 4, 2: 4, 4: 2}.get(num_stages, 1)
     if args.gas == 0:
-        args.gas = max(num_stages * 2, 64 // args.micro_batch_size)
+        # deeper pipelines get more microbatches: bubble = (P-1)/(M+P-1)
+        args.gas = max(num_stages * 16, 64 // args.micro_batch_size)
     # weak scaling: model depth grows with pipeline depth
     num_layers = args.layers_per_stage * num_stages
     mcfg = model_config(args.model, num_layers=num_layers, max_seq_len=args.seq_len)

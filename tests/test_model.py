@@ -68,3 +68,30 @@ def test_gqa_presets():
     ids = torch.randint(0, 128, (2, 32))
     loss = model.compute_loss(ids, ids.clone())
     assert torch.isfinite(loss)
+
+
+def test_selective_checkpointing_equivalence():
+    """Per-layer selective checkpointing must not change the math."""
+    import torch
+    from lpp_amd.config import model_config
+    from lpp_amd.models import LlamaForCausalLM, get_layers_from_config
+
+    mcfg = model_config("llama-tiny", num_layers=4, max_seq_len=32)
+    torch.manual_seed(3)
+    ref = LlamaForCausalLM(mcfg)
+    specs = get_layers_from_config(mcfg, checkpoint_fn=lambda i: i % 2 == 0)
+    torch.manual_seed(3)
+    sel = torch.nn.ModuleList([sp.build() for sp in specs])
+    for a, b in zip(ref.layers.state_dict().values(), sel.state_dict().values()):
+        assert torch.equal(a, b)
+    ids = torch.randint(0, mcfg.vocab_size, (2, 16))
+    ref.train(); x1 = ids
+    for l in ref.layers: x1 = l(x1)
+    x2 = ids
+    for l in sel: l.train(); x2 = l(x2)
+    assert torch.allclose(x1, x2, atol=1e-6)
+    x2.sum().backward()
+    x1.sum().backward()
+    g1 = next(ref.layers[1].parameters()).grad
+    g2 = next(sel[1].parameters()).grad
+    assert torch.allclose(g1, g2, atol=1e-5)
