@@ -169,6 +169,7 @@ class TrainConfig:
 
     # Data
     total_dataset_len: int = 0  # quirk Q3 fix: broadcast once (see engine)
+    data_pattern: str = "uniform"  # synthetic data: uniform | arith (learnable)
     num_workers: int = 2
 
     # Comm

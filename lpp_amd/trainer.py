@@ -42,7 +42,8 @@ def build_dataset(cfg: TrainConfig):
     """Synthetic dataset by default (no-network environment); a real corpus
     drops in behind the same dict contract."""
     n = cfg.total_dataset_len or 4096
-    return SyntheticCausalLMDataset(n, cfg.seq_len, cfg.model.vocab_size, seed=cfg.seed)
+    return SyntheticCausalLMDataset(n, cfg.seq_len, cfg.model.vocab_size, seed=cfg.seed,
+                                    pattern=getattr(cfg, "data_pattern", "uniform"))
 
 
 def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dict:
