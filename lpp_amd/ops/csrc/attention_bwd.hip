@@ -84,22 +84,6 @@ __device__ __forceinline__ void quad_transpose(const bf16x8& v, int st_r, int ou
 // element (d, r) = (d*64 + (r ^ ((((d>>3) ^ d) & 7) << 3))) * 2.  Writes are
 // ~2-way (the XOR varies with the lane's d), B-fragment reads conflict-free
 // (measured 0.9% conflict cycles in the forward vs 8-15% for padded strides).
-__device__ __forceinline__ int tswz32(int d, int r) {
-  return (d * 32 + (r ^ ((((d >> 3) ^ d) & 3) << 3))) * 2;
-}
-__device__ __forceinline__ void write_transposed32(char* img, int r0, int d0,
-                                                   const int dw[4]) {
-#pragma unroll
-  for (int e = 0; e < 2; ++e) {
-    const int d = d0 + e;
-    unsigned w01 = e ? (((unsigned)dw[0] >> 16) | ((unsigned)dw[1] & 0xffff0000u))
-                     : (((unsigned)dw[0] & 0xffffu) | ((unsigned)dw[1] << 16));
-    unsigned w23 = e ? (((unsigned)dw[2] >> 16) | ((unsigned)dw[3] & 0xffff0000u))
-                     : (((unsigned)dw[2] & 0xffffu) | ((unsigned)dw[3] << 16));
-    int2v pair = {(int)w01, (int)w23};
-    *reinterpret_cast<int2v*>(img + tswz32(d, r0)) = pair;
-  }
-}
 __device__ __forceinline__ int tswz(int d, int r) {
   return (d * 64 + (r ^ ((((d >> 3) ^ d) & 7) << 3))) * 2;
 }
@@ -327,28 +311,9 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
 //   dV  += mfma16(pack(P),  dO^T image);  dK += mfma16(pack(dS), Q^T image)
 // GQA: the G query heads sharing a kv head accumulate in-register.
 constexpr int KV_KW = 16, KV_WG = 128, KV_QT = 64;
-constexpr int TS72 = 72;  // 64-row transposed-image stride (odd word count)
 constexpr int TS40 = 40;  // transposed-image row stride (odd word count)
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
-
-// write the quad-transposed columns into a [128][TS72] image.
-// Column order is staggered by lane ((d0>>4)+jj)&1 so each wave store
-// instruction covers both d-parities: 4-way banked instead of 8-way.
-__device__ __forceinline__ void write_transposed72(char* img, int r0, int d0,
-                                                   const int dw[4]) {
-#pragma unroll
-  for (int jj = 0; jj < 2; ++jj) {
-    const int e = ((d0 >> 4) + jj) & 1;
-    const int d = d0 + e;
-    unsigned w01 = e ? (((unsigned)dw[0] >> 16) | ((unsigned)dw[1] & 0xffff0000u))
-                     : (((unsigned)dw[0] & 0xffffu) | ((unsigned)dw[1] << 16));
-    unsigned w23 = e ? (((unsigned)dw[2] >> 16) | ((unsigned)dw[3] & 0xffff0000u))
-                     : (((unsigned)dw[2] & 0xffffu) | ((unsigned)dw[3] << 16));
-    int2v pair = {(int)w01, (int)w23};
-    *reinterpret_cast<int2v*>(img + (d * 72 + r0) * 2) = pair;
-  }
-}
 
 
 // Pack a (lane = n-axis, regs = crow16 q-axis) C-layout value set into the
@@ -390,10 +355,10 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
   // aligned: Qt/dOt are [128][72]*2B = 18432 B each.
   auto q_lds = [&](int buf) -> char* { return smem + buf * 16384; };
   auto do_lds = [&](int buf) -> char* { return smem + 32768 + buf * 16384; };
-  auto qt_lds = [&](int buf) -> char* { return smem + 65536 + buf * 18432; };
-  auto dot_lds = [&](int buf) -> char* { return smem + 102400 + buf * 18432; };
-  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 139264) + buf * 64; };
-  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 139776) + buf * 64; };
+  auto qt_lds = [&](int buf) -> char* { return smem + 65536 + buf * 16384; };
+  auto dot_lds = [&](int buf) -> char* { return smem + 98304 + buf * 16384; };
+  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 131072) + buf * 64; };
+  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 131584) + buf * 64; };
 
 
   const int tid = threadIdx.x;
@@ -451,9 +416,9 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
       *reinterpret_cast<bf16x8*>(do_lds(buf) + bswz(r, st_c)) = dreg;
       int dw[4];
       quad_transpose(qreg, st_r, dw);
-      write_transposed72(qt_lds(buf), r - st_r, st_c + 2 * st_r, dw);
+      write_transposed(qt_lds(buf), r - st_r, st_c + 2 * st_r, dw);
       quad_transpose(dreg, st_r, dw);
-      write_transposed72(dot_lds(buf), r - st_r, st_c + 2 * st_r, dw);
+      write_transposed(dot_lds(buf), r - st_r, st_c + 2 * st_r, dw);
       if (pass == 0 && tid < 128) {
         const int qi = min(qt0 + (tid & 63), S - 1);
         if (tid < 64) l_buf(buf)[tid] = LSE2[ld_base + qi];
@@ -527,9 +492,9 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
           for (int dt = 0; dt < 8; ++dt) {
             const int d = dt * 16 + l16;
             const bf16x8 bd = *reinterpret_cast<const bf16x8*>(
-                dot_lds(cur) + (d * TS72 + half * 32 + hi4 * 8) * 2);
+                dot_lds(cur) + tswz(d, half * 32 + hi4 * 8));
             const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-                qt_lds(cur) + (d * TS72 + half * 32 + hi4 * 8) * 2);
+                qt_lds(cur) + tswz(d, half * 32 + hi4 * 8));
             dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bd, dv_acc[dt], 0, 0, 0);
             dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bq, dk_acc[dt], 0, 0, 0);
           }
@@ -600,7 +565,7 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   }
   {
     const int kvblocks = (S + lpp::KV_WG - 1) / lpp::KV_WG;
-    const size_t lds = 140288;
+    const size_t lds = 132096;
     hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(512), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
