@@ -139,3 +139,42 @@ def test_collator_feeds_engine_loss():
     logits = torch.randn(1, 12, 128)
     loss = loss_fn(logits, out["labels"])
     assert torch.isfinite(loss)
+
+
+def _offline_hf_tokenizer(vocab_size=200):
+    """Build a real HF fast tokenizer entirely offline (no downloads)."""
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+    from transformers import PreTrainedTokenizerFast
+
+    tok = Tokenizer(models.BPE(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.BpeTrainer(
+        vocab_size=vocab_size, special_tokens=["<unk>", "<s>", "</s>"])
+    corpus = ["the cat sat on the mat", "a dog ran fast", "what is two plus two",
+              "four red blue colors", "name a color please"]
+    tok.train_from_iterator(corpus, trainer)
+    t = PreTrainedTokenizerFast(tokenizer_object=tok, unk_token="<unk>",
+                                bos_token="<s>", eos_token="</s>")
+    t.name_or_path = "llama-offline-test"  # routes through the llama branch
+    return t
+
+
+def test_text_collator_with_real_hf_tokenizer():
+    """The collator contract holds with a genuine transformers tokenizer
+    (exercises expand_special_tokenizer's llama path + pad handling on the
+    real API, not just SimpleTokenizer)."""
+    tok = _offline_hf_tokenizer()
+    coll = TextCollator(tok, max_seq_length=12)
+    assert tok.pad_token is not None  # expand added [PAD] or fell back
+    out = coll([
+        {"inputs": "the cat sat", "targets": "on the mat"},
+        {"inputs": "name a color", "targets": "red"},
+    ])
+    assert set(out) == {"input_ids", "labels"}
+    assert out["input_ids"].shape == (2, 12)
+    keep = out["labels"] != IGNORE_INDEX
+    assert keep.any()
+    assert torch.equal(out["labels"][keep], out["input_ids"][keep])
+    # prompt region masked
+    p0 = len(tok("the cat sat")["input_ids"])
+    assert (out["labels"][0, :p0] == IGNORE_INDEX).all()
