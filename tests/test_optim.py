@@ -97,3 +97,50 @@ def test_bf16_params_fp32_master():
     assert opt.masters[0].dtype == torch.float32
     opt.step()
     assert m.weight.dtype == torch.bfloat16
+
+
+def test_dynamic_loss_scaler_schedule():
+    """fp16 scaler: halve after hysteresis overflows, double after window
+    good steps, floor at min_scale (conf/...yaml:137-143)."""
+    from lpp_amd.engine import DynamicLossScaler
+
+    s = DynamicLossScaler(init_scale=2.0**12, scale_window=4, hysteresis=2, min_scale=1.0)
+    assert s.scale == 4096
+    s.update(found_inf=True)       # hysteresis 2 -> first inf tolerated
+    assert s.scale == 4096
+    s.update(found_inf=True)       # second inf -> halve
+    assert s.scale == 2048
+    for _ in range(4):             # window good steps -> double
+        s.update(found_inf=False)
+    assert s.scale == 4096
+    for _ in range(30):            # repeated overflow floors at min
+        s.update(found_inf=True)
+    assert s.scale == 1.0
+
+
+def test_engine_skips_step_on_overflow():
+    """Inf gradient -> skipped_steps increments, params untouched (fp16
+    loss-scale skip semantics; engine._optimizer_step)."""
+    import torch
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.engine import DynamicLossScaler, PipelineEngine
+    from lpp_amd.models import get_layers_from_config, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+
+    mcfg = model_config("llama-tiny", num_layers=1, max_seq_len=32)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=1,
+                      gradient_accumulation_steps=1, seq_len=32, dtype="fp32")
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    engine = PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    engine.loss_scaler = DynamicLossScaler()  # force the fp16 skip path
+    engine.optimizer.flat_grads.fill_(float("inf"))
+    before = [p.clone() for p in module.parameters()]
+    engine._optimizer_step()
+    assert engine.skipped_steps == 1
+    for p0, p1 in zip(before, module.parameters()):
+        assert torch.equal(p0, p1)
+    # grads cleared so the next step starts clean
+    assert engine.optimizer.flat_grads.abs().sum() == 0
