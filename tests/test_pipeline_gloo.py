@@ -81,3 +81,21 @@ def test_eval_batch_pp2():
     got = run_dist(2, _eval_run)
     assert got[0] == pytest.approx(got[1], abs=1e-6)
     assert got[0] > 0
+
+
+def test_pp4_matches_single_process():
+    """4-stage pipeline (deep warmup/cooldown, the N=8 schedule shape in
+    miniature) reproduces the monolithic trajectory."""
+    base = _single_process_baseline(steps=2, gas=8)
+    got = run_dist(4, run_steps, 4, 2, 8)  # stages=4
+    for r in range(4):
+        for a, b in zip(base, got[r]):
+            assert abs(a - b) < 1e-3, (base, got[r])
+
+
+def test_pp4_gas_less_than_stages():
+    """gas=2 < stages=4: some stages do zero steady-state iterations."""
+    base = _single_process_baseline(steps=2, gas=2)
+    got = run_dist(4, run_steps, 4, 2, 2)
+    for a, b in zip(base, got[0]):
+        assert abs(a - b) < 1e-3
