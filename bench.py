@@ -11,9 +11,11 @@ and the ideal scaling curve is FLAT tokens/s (the model grows with N; any
 drop measures pipeline bubble + p2p cost).
 
 Synthetic data (random tokens of the BASELINE shape), random-init weights,
-bf16 compute, fp32 grad accumulation + master AdamW, activation
-checkpointing interval 1 — nothing is skipped inside the timed region: each
-step runs `gas` microbatches forward+backward plus the full optimizer step.
+bf16 compute, fp32 grad accumulation + master AdamW.  Activation
+checkpointing is SELECTIVE: the auto policy recomputes only as many layers
+per stage as the 288 GB memory budget demands (usually zero).  Nothing is
+skipped inside the timed region: each step runs `gas` microbatches
+forward+backward plus the full optimizer step.
 
 Launch (the driver's contract):
   python bench.py --gpus 1 --steps 3 --warmup 1

@@ -13,7 +13,9 @@ and the extension failed to load, we raise instead of silently falling back
 
 The per-op kernel manifest mirrors SURVEY.md §2.7 (the reference's compute
 graph is HF LlamaDecoderLayer + loss_fn; models/llama_ds_mp_wrap.py:8-13,
-105-116 — it ships no kernels of its own).
+105-116 — it ships no kernels of its own): rmsnorm, rope, swiglu, fused
+shifted cross-entropy, fused AdamW, flash causal attention fwd+bwd, and
+the hipBLASLt fp32-accumulating weight-gradient GEMM (linear.py).
 """
 
 from __future__ import annotations
