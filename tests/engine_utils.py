@@ -14,8 +14,10 @@ from lpp_amd.topology import ProcessGrid
 
 
 def make_config(num_stages=1, gas=4, mbs=2, seq=32, dtype="fp32", lr=1e-3):
+    # 8 decoder layers: every stage owns >=1 spec up to PP=8, and the model
+    # is identical regardless of the stage count (trajectory comparisons)
     cfg = TrainConfig(
-        model=model_config("llama-tiny"),
+        model=model_config("llama-tiny", num_layers=8),
         num_stages=num_stages,
         micro_batch_size=mbs,
         gradient_accumulation_steps=gas,

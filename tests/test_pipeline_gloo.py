@@ -99,3 +99,12 @@ def test_pp4_gas_less_than_stages():
     got = run_dist(4, run_steps, 4, 2, 2)
     for a, b in zip(base, got[0]):
         assert abs(a - b) < 1e-3
+
+
+@pytest.mark.slow
+def test_pp8_matches_single_process():
+    """world 8 = the driver's N=8 scale-run schedule shape, on gloo."""
+    base = _single_process_baseline(steps=2, gas=8)
+    got = run_dist(8, run_steps, 8, 2, 8, timeout=600.0)
+    for a, b in zip(base, got[0]):
+        assert abs(a - b) < 1e-3, (base, got[0])
