@@ -114,6 +114,35 @@ def bench_ce(rows=4096, V=32000):
     return rows_out
 
 
+def bench_attention(B=1, S=4096, H=64, D=128):
+    """Flash fwd/bwd vs torch SDPA (aotriton) at the 65B shape."""
+    from lpp_amd import ops
+    from lpp_amd.ops.attention import causal_attention_ref
+
+    ext = ops.extension()
+    q = torch.randn(B, S, H, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    do = torch.randn_like(q)
+    flops_fwd = 4 * B * H * S * S * D / 2
+    t_hip = timeit(lambda: ext.attention_fwd(q, k, v))
+    t_ref = timeit(lambda: causal_attention_ref(q, k, v))
+    rows = [("attn_fwd", t_hip, t_ref, int(flops_fwd))]
+
+    o, lse2 = ext.attention_fwd(q, k, v)
+    t_hipb = timeit(lambda: ext.attention_bwd(do, q, k, v, o, lse2))
+
+    def refb():
+        qr = q.detach().requires_grad_(True)
+        kr = k.detach().requires_grad_(True)
+        vr = v.detach().requires_grad_(True)
+        causal_attention_ref(qr, kr, vr).backward(do)
+
+    t_refb = timeit(refb, iters=5, warmup=2)
+    rows.append(("attn_bwd", t_hipb, t_refb, int(flops_fwd * 2.5)))
+    return rows
+
+
 def bench_adamw(n=1_000_000_000 // 4):
     from lpp_amd import ops
 
@@ -144,31 +173,6 @@ def bench_adamw(n=1_000_000_000 // 4):
 
     t_ref = timeit(ref, iters=10)
     return [("fused_adamw_256M", t_hip, t_ref, bytes_moved)]
-
-
-def bench_attention(B=1, S=4096, H=64, D=128):
-    from lpp_amd.ops.attention import causal_attention
-
-    q = torch.randn(B, S, H, D, device=DEV, dtype=torch.bfloat16)
-    k = torch.randn_like(q)
-    v = torch.randn_like(q)
-    flops = 4 * B * H * S * S * D / 2  # causal halves the work
-    t = timeit(lambda: causal_attention(q, k, v))
-    out = [("attention_fwd(dispatch)", t, t, 0)]
-    qt, kt, vt = (x.transpose(1, 2) for x in (q, k, v))
-    from torch.nn.attention import SDPBackend, sdpa_kernel
-
-    for name, be in [("flash", SDPBackend.FLASH_ATTENTION),
-                     ("mem_eff", SDPBackend.EFFICIENT_ATTENTION),
-                     ("math", SDPBackend.MATH)]:
-        try:
-            with sdpa_kernel(be):
-                tt = timeit(lambda: torch.nn.functional.scaled_dot_product_attention(
-                    qt, kt, vt, is_causal=True))
-            print(f"  sdpa[{name}]: {tt:.3f} ms = {flops / tt / 1e9:.0f} TF/s")
-        except Exception as e:
-            print(f"  sdpa[{name}]: unavailable ({type(e).__name__})")
-    return out
 
 
 def main():
