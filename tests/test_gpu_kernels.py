@@ -227,3 +227,15 @@ def test_lp_linear_fused_wgrad_matches_autograd():
     assert err < 2e-2, err
     err_x = (x.grad.float() - xr.grad.float()).abs().max()
     assert err_x < 1e-2, err_x
+
+
+def test_rmsnorm_bwd_deterministic_dw(ext):
+    """The partial-buffer dw reduction is order-deterministic (no atomics)."""
+    torch.manual_seed(13)
+    x = torch.randn(4096, 8192, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(8192, device="cuda", dtype=torch.float32)
+    dy = torch.randn_like(x)
+    _, inv = ext.rmsnorm_fwd(x, w, 1e-6)
+    _, dw1 = ext.rmsnorm_bwd(dy, x, w, inv)
+    _, dw2 = ext.rmsnorm_bwd(dy, x, w, inv)
+    assert torch.equal(dw1, dw2)
