@@ -51,3 +51,21 @@ def test_llama_spec_weights():
     bounds = partition_balanced(weights, 8)
     parts = [sum(weights[bounds[s] : bounds[s + 1]]) for s in range(8)]
     assert max(parts) / min(parts) < 1.35  # reasonably balanced for 83 layers / 8 stages
+
+
+def test_65b_pp8_partition_balanced():
+    """The headline N=8 partition: embed rides with stage 0, norm+head with
+    stage 7, decoder layers 10-11 per stage, param spread < 4%."""
+    from lpp_amd.config import model_config
+    from lpp_amd.layer_spec import partition_balanced
+    from lpp_amd.models import get_layers_from_config
+
+    m = model_config("llama-65b", num_layers=80, max_seq_len=4096)
+    specs = get_layers_from_config(m)
+    w = [s.param_count() for s in specs]
+    bounds = partition_balanced(w, 8)
+    per_stage = [sum(w[bounds[i]:bounds[i + 1]]) for i in range(8)]
+    assert bounds[0] == 0 and bounds[-1] == len(specs) == 83
+    assert max(per_stage) / min(per_stage) < 1.04
+    n_specs = [bounds[i + 1] - bounds[i] for i in range(8)]
+    assert all(10 <= n <= 12 for n in n_specs)
