@@ -176,6 +176,7 @@ def main() -> int:
 
     for _ in range(args.warmup):
         engine.train_batch(it)
+    engine.timer_summary(reset=True)  # timers cover only the timed steps
 
     sync()
     t0 = time.time()
@@ -195,6 +196,23 @@ def main() -> int:
     tokens_per_step = args.gas * args.micro_batch_size * args.seq_len * dp
     tokens_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
+
+    # per-rank section timers (host-side, seconds over the timed steps) so a
+    # multi-GPU run's bubble/p2p/compute split is diagnosable from the JSON
+    tsum = engine.timer_summary()
+    tvec = torch.tensor([tsum["forward"], tsum["backward"], tsum["p2p"],
+                         tsum["allreduce"], tsum["optimizer"]], dtype=torch.float64)
+    if dist.is_initialized():
+        gathered = [torch.zeros_like(tvec) for _ in range(world)]
+        dist.all_gather(gathered, tvec)
+    else:
+        gathered = [tvec]
+    stage_timers = [
+        {"rank": i, "fwd_s": round(float(g[0]), 2), "bwd_s": round(float(g[1]), 2),
+         "p2p_s": round(float(g[2]), 2), "allreduce_s": round(float(g[3]), 2),
+         "optim_s": round(float(g[4]), 2)}
+        for i, g in enumerate(gathered)
+    ]
 
     if rank == 0:
         par = f"pp{num_stages}" + (f"_dp{dp}" if dp > 1 else "")
@@ -225,6 +243,7 @@ def main() -> int:
                 "peak_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2)
                 if on_gpu
                 else None,
+                "stage_timers": stage_timers,
             },
         }
         print(json.dumps(result), flush=True)
