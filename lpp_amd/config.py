@@ -170,6 +170,9 @@ class TrainConfig:
     # Data
     total_dataset_len: int = 0  # quirk Q3 fix: broadcast once (see engine)
     data_pattern: str = "uniform"  # synthetic data: uniform | arith (learnable)
+    # shell command run by rank 0 after each checkpoint save, with {dir}
+    # substituted (reference: ./s5cmd sync to S3, trainer_base_ds_mp.py:220)
+    save_hook_cmd: str = ""
     num_workers: int = 2
 
     # Comm

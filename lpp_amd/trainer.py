@@ -134,6 +134,15 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
                     engine, cfg.output_dir, tag=f"global_step{step}",
                     client_state={"step": step},
                 )
+                if rank0 and cfg.save_hook_cmd:
+                    # reference parity: post-save sync hook (s5cmd, :220)
+                    import subprocess
+
+                    cmd = cfg.save_hook_cmd.format(
+                        dir=os.path.join(cfg.output_dir, f"global_step{step}"))
+                    r = subprocess.run(cmd, shell=True)
+                    if r.returncode != 0:
+                        logger.warning("save_hook_cmd failed (%d): %s", r.returncode, cmd)
     final = {"steps": step, "elapsed": time.time() - t_start}
     if cfg.save_steps:
         save_engine_checkpoint(engine, cfg.output_dir, tag=f"global_step{step}",

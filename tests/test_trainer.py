@@ -56,3 +56,23 @@ def test_trainer_resume(tmp_path):
 def test_trainer_cli_overrides(tmp_path):
     cfg_path = _write_cfg(tmp_path, max_steps=1, save_steps=0)
     assert trainer_main(["--config", cfg_path, "optimizer.lr=5e-4", "max_steps=2"]) == 0
+
+
+def test_save_hook_cmd(tmp_path):
+    """Post-save shell hook runs with {dir} substituted (reference s5cmd
+    sync slot, trainer_base_ds_mp.py:220)."""
+    from lpp_amd.trainer import main
+
+    marker = tmp_path / "hook_ran"
+    out = tmp_path / "out"
+    rc = main([
+        "--config", "conf/llama_7b_pp2_cpu.yaml",
+        "num_stages=1", "model.name=llama-tiny", "model.num_layers=2",
+        "micro_batch_size=2", "gradient_accumulation_steps=2", "seq_len=32",
+        "max_steps=2", "save_steps=2", "logging_steps=1", "total_dataset_len=16",
+        f"output_dir={out}",
+        f"save_hook_cmd=echo {{dir}} > {marker}",
+    ])
+    assert rc == 0
+    assert marker.exists()
+    assert "global_step2" in marker.read_text()
