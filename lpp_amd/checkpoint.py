@@ -142,10 +142,9 @@ def load_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
         raise FileNotFoundError(f"no 'latest' tag in {ckpt_dir}")
     load_module_weights(engine.module, ckpt_dir, tag, dtype=engine.dtype)
     engine.module.to(engine.device)
-    # refresh masters from (re)loaded params
-    with torch.no_grad():
-        for p, m in zip(engine.optimizer.params, engine.optimizer.masters):
-            m.copy_(p.detach().float())
+    # refresh masters from (re)loaded params (mode-aware: ZeRO-1 keeps one
+    # flat master shard, not per-param copies)
+    engine.optimizer.refresh_masters()
 
     client_state = None
     if not load_module_only:

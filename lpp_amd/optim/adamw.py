@@ -123,6 +123,16 @@ class MixedPrecisionAdamW:
     def is_sharded(self) -> bool:
         return self.shard_world > 1
 
+    @torch.no_grad()
+    def refresh_masters(self) -> None:
+        """Re-derive fp32 master state from the (re)loaded model params —
+        used after a module-only warm start."""
+        if self.is_sharded:
+            self.masters[0].copy_(self.param_shard.to(torch.float32))
+        else:
+            for p, m in zip(self.params, self.masters):
+                m.copy_(p.detach().to(torch.float32))
+
     def grad_sq_sum(self) -> torch.Tensor:
         """Local sum of squared gradients (fp32 scalar tensor).  Unsharded:
         the full flat buffer (callers all-reduce across the PIPE group).
