@@ -37,7 +37,7 @@ import torch.distributed as dist
 from lpp_amd.config import TrainConfig, model_config
 from lpp_amd.data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset
 from lpp_amd.engine import PipelineEngine
-from lpp_amd.models import DecoderLayerPipe, EmbeddingPipe, RMSNorm, get_layers_from_config, loss_fn
+from lpp_amd.models import RMSNorm, get_layers_from_config, loss_fn
 from lpp_amd.pipeline_module import PipelineModule
 from lpp_amd.topology import ProcessGrid
 from lpp_amd.utils import init_distributed, set_seed

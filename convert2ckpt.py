@@ -27,7 +27,6 @@ reference resizes after expand_special_tokenizer, convert2ckpt.py:59-63).
 from __future__ import annotations
 
 import argparse
-import json
 import re
 from pathlib import Path
 

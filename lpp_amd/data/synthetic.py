@@ -20,7 +20,7 @@ weights, synthetic data of the reference's shape).
 
 from __future__ import annotations
 
-from typing import Dict, Iterator, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 from torch.utils.data import DataLoader, Dataset, DistributedSampler, RandomSampler

@@ -13,7 +13,7 @@ contiguous slices minimising the maximum per-stage parameter count
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import List, Sequence
 
 
 class LayerSpec:

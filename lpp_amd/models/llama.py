@@ -22,7 +22,6 @@ State-dict keys match HF ``LlamaForCausalLM`` per-layer keys with the
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional
 
 import torch
