@@ -43,6 +43,26 @@ class RMSNorm(nn.Module):
         return ops.rmsnorm(x, self.weight, self.variance_epsilon)
 
 
+class KVCache:
+    """Per-layer key/value cache for autoregressive decoding
+    ([B, max_len, Hkv, D] bf16/fp32; ``length`` = filled prefix).  The
+    reference is a training template with no inference path; this serves
+    the framework's deployment story (single-device generation)."""
+
+    def __init__(self, batch: int, max_len: int, kv_heads: int, head_dim: int,
+                 device, dtype):
+        self.k = torch.zeros(batch, max_len, kv_heads, head_dim, device=device, dtype=dtype)
+        self.v = torch.zeros_like(self.k)
+        self.length = 0
+
+    def append(self, k: torch.Tensor, v: torch.Tensor) -> int:
+        S = k.shape[1]
+        self.k[:, self.length : self.length + S] = k
+        self.v[:, self.length : self.length + S] = v
+        self.length += S
+        return self.length
+
+
 class LlamaAttention(nn.Module):
     """Self-attention with RoPE; q/k/v/o projections named for HF key parity."""
 
@@ -59,7 +79,7 @@ class LlamaAttention(nn.Module):
         self.v_proj = LPLinear(cfg.hidden_size, self.num_kv_heads * self.head_dim)
         self.o_proj = LPLinear(self.num_heads * self.head_dim, cfg.hidden_size)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, cache: Optional["KVCache"] = None) -> torch.Tensor:
         B, S, _ = x.shape
         q = self.q_proj(x).view(B, S, self.num_heads, self.head_dim)
         k = self.k_proj(x).view(B, S, self.num_kv_heads, self.head_dim)
@@ -67,9 +87,37 @@ class LlamaAttention(nn.Module):
         cos, sin = ops.build_rope_cache(
             self.max_seq_len, self.head_dim, self.rope_theta, x.device
         )
-        q = ops.apply_rope(q, cos, sin)
-        k = ops.apply_rope(k, cos, sin)
-        o = ops.causal_attention(q, k, v)  # [B,S,H,D]
+        pos = cache.length if cache is not None else 0
+        q = ops.apply_rope(q, cos, sin, pos_offset=pos)
+        k = ops.apply_rope(k, cos, sin, pos_offset=pos)
+        if cache is None:
+            o = ops.causal_attention(q, k, v)  # [B,S,H,D]
+        else:
+            total = cache.append(k, v)
+            kc = cache.k[:, :total]
+            vc = cache.v[:, :total]
+            if S == total:
+                # prefill: plain causal over the whole prefix (flash path)
+                o = ops.causal_attention(q, k, v)
+            else:
+                # decode: q attends the full cached prefix.  rows see
+                # positions <= their own: causal offset mask for S > 1,
+                # no mask needed for single-token decode.
+                rep = self.num_heads // self.num_kv_heads
+                qt = q.transpose(1, 2)
+                kt = kc.transpose(1, 2)
+                vt = vc.transpose(1, 2)
+                if rep > 1:
+                    kt = kt.repeat_interleave(rep, dim=1)
+                    vt = vt.repeat_interleave(rep, dim=1)
+                mask = None
+                if S > 1:
+                    qpos = torch.arange(pos, total, device=x.device)
+                    kpos = torch.arange(total, device=x.device)
+                    mask = kpos[None, :] <= qpos[:, None]
+                o = torch.nn.functional.scaled_dot_product_attention(
+                    qt, kt, vt, attn_mask=mask
+                ).transpose(1, 2).contiguous()
         return self.o_proj(o.reshape(B, S, self.num_heads * self.head_dim))
 
 
@@ -100,17 +148,19 @@ class DecoderLayerPipe(nn.Module):
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
         self.activation_checkpointing = activation_checkpointing
 
-    def _forward_impl(self, hidden: torch.Tensor) -> torch.Tensor:
-        hidden = hidden + self.self_attn(self.input_layernorm(hidden))
+    def _forward_impl(self, hidden: torch.Tensor,
+                      cache: Optional[KVCache] = None) -> torch.Tensor:
+        hidden = hidden + self.self_attn(self.input_layernorm(hidden), cache=cache)
         hidden = hidden + self.mlp(self.post_attention_layernorm(hidden))
         return hidden
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+    def forward(self, hidden: torch.Tensor,
+                cache: Optional[KVCache] = None) -> torch.Tensor:
         if self.activation_checkpointing and self.training and hidden.requires_grad:
             return torch.utils.checkpoint.checkpoint(
                 self._forward_impl, hidden, use_reentrant=False, preserve_rng_state=False
             )
-        return self._forward_impl(hidden)
+        return self._forward_impl(hidden, cache=cache)
 
     @staticmethod
     def spec_param_count(cfg: ModelConfig, activation_checkpointing: bool = False) -> int:
@@ -250,3 +300,53 @@ class LlamaForCausalLM(nn.Module):
 
     def compute_loss(self, input_ids: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         return loss_fn(self.forward(input_ids), labels)
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int,
+                 eos_token_id: Optional[int] = None,
+                 temperature: float = 0.0,
+                 generator: Optional[torch.Generator] = None) -> torch.Tensor:
+        """KV-cached autoregressive generation (greedy, or sampling with
+        ``temperature`` > 0).  Single-device serving path; the reference has
+        no inference capability at all (it is a training template)."""
+        was_training = self.training
+        self.eval()
+        B, S0 = input_ids.shape
+        dev = input_ids.device
+        p = next(self.parameters())
+        max_len = S0 + max_new_tokens
+        caches = [
+            KVCache(B, max_len, self.cfg.kv_heads, self.cfg.head_dim, dev, p.dtype)
+            for _ in range(self.cfg.num_layers)
+        ]
+
+        def step(ids: torch.Tensor) -> torch.Tensor:
+            x = ids
+            li = 0
+            for layer in self.layers:
+                if isinstance(layer, DecoderLayerPipe):
+                    x = layer(x, cache=caches[li])
+                    li += 1
+                else:
+                    x = layer(x)
+            return x[:, -1]  # last-position logits [B, V]
+
+        out = input_ids
+        logits = step(input_ids)
+        finished = torch.zeros(B, dtype=torch.bool, device=dev)
+        for _ in range(max_new_tokens):
+            if temperature > 0:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                nxt = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+            else:
+                nxt = logits.argmax(dim=-1)
+            if eos_token_id is not None:
+                nxt = torch.where(finished, torch.full_like(nxt, eos_token_id), nxt)
+                finished |= nxt == eos_token_id
+            out = torch.cat([out, nxt[:, None]], dim=1)
+            if eos_token_id is not None and bool(finished.all()):
+                break
+            logits = step(nxt[:, None])
+        if was_training:
+            self.train()
+        return out

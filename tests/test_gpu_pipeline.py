@@ -53,3 +53,26 @@ def test_native_extension_required_on_gpu():
     assert ops.use_hip(x)
     y = ops.rmsnorm(x, w, 1e-6)
     assert y.is_cuda
+
+
+def test_generate_on_gpu():
+    """KV-cached generation runs on the bf16 flash path; the first decoded
+    token equals the full-forward argmax (identical prefill path)."""
+    import torch
+
+    from lpp_amd.config import model_config
+    from lpp_amd.models import LlamaForCausalLM
+
+    mcfg = model_config("llama-65b", num_layers=2, max_seq_len=256)
+    torch.manual_seed(4)
+    m = LlamaForCausalLM(mcfg).to("cuda", torch.bfloat16)
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, 0.02)
+    ids = torch.randint(0, mcfg.vocab_size, (1, 32), device="cuda")
+    out = m.generate(ids, max_new_tokens=4)
+    assert out.shape == (1, 36)
+    with torch.no_grad():
+        ref_next = m(ids)[:, -1].argmax(-1)
+    assert int(out[0, 32]) == int(ref_next)
