@@ -20,13 +20,16 @@
 //        dP^T = mfma(V_lds, dO_reg)  same layout
 //        P    = exp2(c*S^T - L[own q]);  dS = P*(dP^T - delta[own q])
 //        dQ  += mfma(pack(dS), K^T image)          (scale at epilogue)
-//   3. dK/dV kernel — 4 waves x 32 kv rows (lane = kv), 2 waves/SIMD,
-//      grid over KV blocks x B*HKV; per 64-row q tile (two 32-row halves):
+//   3. dK/dV kernel — 8 waves x 16 kv rows (lane = kv, 16x16x32
+//      fragments), 2 waves/SIMD; grid over KV blocks x B*HKV; per
+//      double-buffered 64-row q tile (two 32-row compute halves):
 //        S    = mfma(Q_lds, K_reg)   C[q regs][kv lane]
 //        dP   = mfma(dO_lds, V_reg)  same layout
 //        P    = exp2(c*S - L[q])  (L,delta broadcast from a staged tile)
 //        dV  += mfma(pack(P),  dO^T image)
 //        dK  += mfma(pack(dS), Q^T image)
+//      P/dS pack to A-fragments via a cvt_pk + permlane32/16_swap network
+//      (pack_frag16) — no LDS roundtrip.
 //      GQA: the G query heads sharing a kv head accumulate in-register.
 //
 // LDS images: "normal" [rows][128] tiles XOR-swizzled ((row&15)<<4) for the
