@@ -1,5 +1,6 @@
 from .llama import (
     DecoderLayerPipe,
+    KVCache,
     EmbeddingPipe,
     LMHeadPipe,
     LlamaForCausalLM,
@@ -15,6 +16,7 @@ from .llama import (
 
 __all__ = [
     "DecoderLayerPipe",
+    "KVCache",
     "EmbeddingPipe",
     "LMHeadPipe",
     "LlamaForCausalLM",
