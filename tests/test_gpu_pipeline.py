@@ -76,3 +76,43 @@ def test_generate_on_gpu():
     with torch.no_grad():
         ref_next = m(ids)[:, -1].argmax(-1)
     assert int(out[0, 32]) == int(ref_next)
+
+
+def test_pipeline_generate_single_stage_gpu():
+    """pipeline_generate on a 1-stage grid, bf16 flash prefill: first
+    decoded token equals the module's full-forward argmax; decode rate
+    printed for the serving evidence."""
+    import time
+
+    import torch
+
+    from lpp_amd.config import model_config
+    from lpp_amd.inference import pipeline_generate
+    from lpp_amd.models import get_layers_from_config, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+
+    mcfg = model_config("llama-65b", num_layers=4, max_seq_len=512)
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cuda"), dtype=torch.bfloat16)
+    with torch.no_grad():
+        for p in module.parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, 0.02)
+    ids = torch.randint(0, mcfg.vocab_size, (1, 128), device="cuda")
+    out = pipeline_generate(module, grid, ids, max_new_tokens=8)
+    assert out.shape == (1, 136)
+    with torch.no_grad():
+        x = ids
+        for layer in module.layers:
+            x = layer(x)
+        ref = x[:, -1].argmax(-1)
+    assert int(out[0, 128]) == int(ref)
+    # decode-rate probe (4 of 80 layers -> scale mentally by 20x)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    out = pipeline_generate(module, grid, ids, max_new_tokens=32)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"\ndecode: {32 / dt:.1f} tok/s on 4/80 of a 65B (B=1, prefill 128)")
