@@ -85,6 +85,20 @@ def _pad_rows(t: torch.Tensor, rows: int) -> torch.Tensor:
     return torch.cat([t, pad], dim=0)
 
 
+def copy_tokenizer_and_config(hf_dir: Path, outdir: Path) -> None:
+    """Pass tokenizer/config files through to the checkpoint dir, as the
+    reference converter does (convert2ckpt.py:79-80) so a training run can
+    point model_name_or_path at one directory."""
+    import shutil
+
+    for name in ("config.json", "generation_config.json", "tokenizer.json",
+                 "tokenizer.model", "tokenizer_config.json",
+                 "special_tokens_map.json"):
+        f = hf_dir / name
+        if f.exists():
+            shutil.copy2(f, outdir / name)
+
+
 def convert_hf(hf_dir: Path, outdir: Path, pad_vocab_to: int, dtype) -> None:
     layer_re = re.compile(r"^model\.layers\.(\d+)\.(.+)$")
     collect: dict[int, dict] = {}
@@ -108,6 +122,7 @@ def convert_hf(hf_dir: Path, outdir: Path, pad_vocab_to: int, dtype) -> None:
     collect[num_layers + 1] = collect.pop("norm")
     collect[num_layers + 2] = collect.pop("head")
     write_ckpt(outdir, num_layers, collect, pad_vocab_to, dtype)
+    copy_tokenizer_and_config(hf_dir, outdir)
     print(f"wrote {num_layers + 3} layer files to {outdir}")
 
 
