@@ -60,7 +60,7 @@ def _ref_attn(q, k, v):
     return (scores.softmax(-1) @ vt).transpose(1, 2)
 
 
-@pytest.mark.parametrize("S", [64, 96, 128, 256, 384, 2048])
+@pytest.mark.parametrize("S", [1, 8, 64, 96, 128, 256, 384, 2048])
 def test_attention_fwd_numerics(ext, S):
     torch.manual_seed(1)
     B, H, D = 2, 4, 128
