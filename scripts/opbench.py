@@ -1,6 +1,4 @@
 #!/usr/bin/env python
-import sys, pathlib
-sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
 """Per-op A/B microbench: HIP kernel vs eager PyTorch on MI355X.
 
 Shapes are the 65B hot-path shapes (hidden 8192, intermediate 22016,
@@ -13,9 +11,13 @@ Run on the GPU box:  python scripts/opbench.py [--csv out.csv]
 from __future__ import annotations
 
 import argparse
+import pathlib
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
 
 DEV = torch.device("cuda", 0)
 
