@@ -117,7 +117,9 @@ def bench_ce(rows=4096, V=32000):
 
 
 def bench_attention(B=1, S=4096, H=64, D=128):
-    """Flash fwd/bwd vs torch SDPA (aotriton) at the 65B shape."""
+    """Flash fwd/bwd vs torch SDPA at the 65B shape.  NOTE: SDPA on ROCm is
+    aotriton's own flash kernel — a much stronger baseline than the
+    materialised-mask eager attention the reference actually runs."""
     from lpp_amd import ops
     from lpp_amd.ops.attention import causal_attention_ref
 
