@@ -163,11 +163,6 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[dt][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;
 
-  // static wave priority for the younger half (guide T5 static form):
-  // must be wave-uniform via readfirstlane or s_setprio applies to all waves
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
-
   // ---- prologue: stage tile 0 ----
   bf16x8 kreg0, kreg1, vreg0, vreg1;
   ld_tile(0, 0, kreg0, vreg0);
@@ -192,6 +187,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
       for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
         for (int r = 0; r < 16; ++r) st[t2][r] = 0.f;
+      __builtin_amdgcn_s_setprio(1);  // favour the MFMA cluster (guide T5)
 #pragma unroll
       for (int dc = 0; dc < 8; ++dc) {
 #pragma unroll
@@ -201,6 +197,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
           st[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dc], st[t2], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- causal / tail mask -> ms (unscaled scores, -inf where masked) ----
       // lane's q row = qrow; score reg r of tile t2 is kv
@@ -278,6 +275,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
         }
 
       // ---- O += P V  (B-frags from the transposed V image) ----
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
         const int d = dt * 32 + lq;
@@ -288,6 +286,7 @@ __global__ __launch_bounds__(512, 2) void attn_fwd_kernel(
           o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], bv, o_acc[dt], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     // ---- stage next tile; one barrier per tile ----
