@@ -38,3 +38,21 @@ def test_unknown_key_rejected():
 def test_global_batch():
     cfg = TrainConfig(micro_batch_size=8, gradient_accumulation_steps=256)
     assert cfg.global_batch_size(dp_degree=2) == 4096  # reference derived batch
+
+
+def test_all_shipped_configs_load():
+    """Every conf/*.yaml parses onto the dataclasses and is self-consistent."""
+    import glob
+
+    from lpp_amd.config import TrainConfig, model_config
+
+    files = sorted(glob.glob("conf/*.yaml"))
+    assert len(files) >= 6
+    for f in files:
+        cfg = TrainConfig.load(f)
+        m = cfg.model
+        assert m.hidden_size % m.num_heads == 0
+        assert cfg.seq_len <= m.max_seq_len
+        assert cfg.num_stages >= 1
+        # named preset exists and matches the file's geometry family
+        model_config(m.name)
