@@ -94,33 +94,19 @@ def main() -> int:
 
     num_stages = world // args.dp if world > 1 else 1
     dp = args.dp if world > 1 else 1
-    if args.micro_batch_size == 0:
-        # larger microbatches raise GEMM/kernel efficiency (~+8% at mbs 4)
-        # but multiply in-flight activation memory by the pipeline depth and
-        # shrink the microbatch count (pipeline bubble) at fixed tokens/step
-        args.micro_batch_size = {1: 4, 2: 4, 4: 2}.get(num_stages, 1)
-    if args.gas == 0:
-        # deeper pipelines get more microbatches: bubble = (P-1)/(M+P-1)
-        args.gas = max(num_stages * 16, 64 // args.micro_batch_size)
     # weak scaling: model depth grows with pipeline depth
     num_layers = args.layers_per_stage * num_stages
     mcfg = model_config(args.model, num_layers=num_layers, max_seq_len=args.seq_len)
 
-    # MI355X-first memory policy: the reference NEEDS activation
-    # checkpointing on 80 GB A100s (conf/...yaml:19); on 288 GB HBM3E most
-    # or all of a 65B stage's activations fit, so recompute only as many
-    # layers per stage as the budget demands (selective checkpointing;
-    # full recompute costs ~+33% forward work).
+    # MI355X-first schedule/memory policy (lpp_amd.utils.schedule)
+    from lpp_amd.utils.schedule import choose_schedule
+
+    sched = choose_schedule(mcfg, num_stages, args.layers_per_stage, args.seq_len,
+                            args.micro_batch_size, args.gas)
+    args.micro_batch_size = sched.micro_batch_size
+    args.gas = sched.gas
     if args.act_ckpt == "auto":
-        S, H, I = args.seq_len, mcfg.hidden_size, mcfg.intermediate_size
-        # saved per non-checkpointed layer (bf16): ~8 S*H tensors + 3 S*I
-        act_per_layer = (8 * S * H + 3 * S * I) * args.micro_batch_size * 2
-        in_flight = min(num_stages, args.gas)  # stage 0 holds the most
-        stage_params = mcfg.num_params() // max(num_stages, 1) + 2 * mcfg.vocab_size * H
-        # bf16 param + fp32 master/m/v/grad = 18 B per param
-        budget = 248e9  # estimator validated vs measured peaks (+-1 GB at N=1)
-        free_layers = max(0, int((budget - stage_params * 18) // (act_per_layer * in_flight)))
-        ckpt_per_stage = max(0, args.layers_per_stage - free_layers)
+        ckpt_per_stage = sched.ckpt_layers_per_stage
     elif args.act_ckpt == "1":
         ckpt_per_stage = args.layers_per_stage
     else:
