@@ -242,6 +242,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
       for (int t2 = 0; t2 < 2; ++t2)
 #pragma unroll
         for (int r = 0; r < 16; ++r) { st[t2][r] = 0.f; dpt[t2][r] = 0.f; }
+      __builtin_amdgcn_s_setprio(1);  // favour the MFMA cluster (guide T5)
 #pragma unroll
       for (int dc = 0; dc < 8; ++dc) {
 #pragma unroll
@@ -254,6 +255,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
           dpt[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[dc], dpt[t2], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       const bool need_mask = (kv0 + DQ_KVB > qw0) || (kv0 + DQ_KVB > S);
       float ds[32];
@@ -274,6 +276,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
       pack_pair(&ds[0], &dsa[0]);
       pack_pair(&ds[16], &dsa[2]);
 
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
         const int d = dt * 32 + lq;
@@ -284,6 +287,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
           dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa[ks], bk, dq_acc[dt], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     if (have_next) {
@@ -429,8 +433,6 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
       }
     };
 
-    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-      __builtin_amdgcn_s_setprio(1);
     bf16x8 qreg0, dreg0, qreg1, dreg1;
     ld_tile(qstart, 0, qreg0, dreg0);
     ld_tile(qstart, 1, qreg1, dreg1);
@@ -455,6 +457,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
           for (int qs = 0; qs < 2; ++qs)
 #pragma unroll
             for (int r = 0; r < 4; ++r) { st[qs][r] = 0.f; dpt[qs][r] = 0.f; }
+          __builtin_amdgcn_s_setprio(1);
 #pragma unroll
           for (int dc = 0; dc < 4; ++dc) {
 #pragma unroll
@@ -467,6 +470,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
               dpt[qs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ad, vf[dc], dpt[qs], 0, 0, 0);
             }
           }
+          __builtin_amdgcn_s_setprio(0);
 
           const bool need_mask = (qh0 < kvw0 + KV_KW) || (qh0 + 32 > S);
           float p[2][4], ds[2][4];
@@ -491,6 +495,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
           const bf16x8 pa = pack_frag16(p, hi4 & 1);
           const bf16x8 dsa = pack_frag16(ds, hi4 & 1);
 
+          __builtin_amdgcn_s_setprio(1);
 #pragma unroll
           for (int dt = 0; dt < 8; ++dt) {
             const int d = dt * 16 + l16;
@@ -501,6 +506,7 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
             dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bd, dv_acc[dt], 0, 0, 0);
             dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, bq, dk_acc[dt], 0, 0, 0);
           }
+          __builtin_amdgcn_s_setprio(0);
         }
       }
 

@@ -88,3 +88,13 @@ def test_partition_rejects_too_many_stages():
 
     with pytest.raises(ValueError):
         partition_balanced([1, 1, 1], 4)
+
+
+def test_step_timer():
+    from lpp_amd.utils import StepTimer
+
+    t = StepTimer()
+    t.start()
+    time.sleep(0.01)
+    dt = t.stop()
+    assert dt >= 0.01 and t.mean >= 0.01
