@@ -280,6 +280,17 @@ def init_monolithic_weights(model: "LlamaForCausalLM", seed: int) -> None:
         deterministic_layer_init(layer, model.cfg, seed, gidx)
 
 
+def layers_from_model(model: "LlamaForCausalLM") -> List[nn.Module]:
+    """Family-A construction (reference ``get_model``,
+    models/llama_ds_mp_wrap.py:119-125): take an ALREADY-LOADED monolithic
+    model and return its flat layer list for a PipelineModule — each rank
+    keeps references (no copy), so like the reference this materialises the
+    full model per worker first (the drawback README.md:21 calls out; the
+    recommended path is ``get_layers_from_config`` + per-stage checkpoint
+    load).  No weight tying, by design (README.md:44-46)."""
+    return list(model.layers)
+
+
 class LlamaForCausalLM(nn.Module):
     """Monolithic (non-pipeline) model built from the same layer specs —
     the numerics oracle for PP-vs-single-process loss-equivalence tests

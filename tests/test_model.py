@@ -95,3 +95,26 @@ def test_selective_checkpointing_equivalence():
     g1 = next(ref.layers[1].parameters()).grad
     g2 = next(sel[1].parameters()).grad
     assert torch.allclose(g1, g2, atol=1e-5)
+
+
+def test_family_a_layers_from_model():
+    """Reference family A (get_model :119-125): pipeline layers taken from a
+    loaded monolithic model share storage and reproduce its forward."""
+    import torch
+
+    from lpp_amd.config import model_config
+    from lpp_amd.models import LlamaForCausalLM, layers_from_model, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+
+    mcfg = model_config("llama-tiny", num_layers=2, max_seq_len=32)
+    torch.manual_seed(8)
+    mono = LlamaForCausalLM(mcfg)
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(layers_from_model(mono), grid, loss_fn=loss_fn)
+    ids = torch.randint(0, mcfg.vocab_size, (1, 16))
+    with torch.no_grad():
+        assert torch.equal(module(ids), mono(ids))
+    # shared storage, not copies
+    assert next(module.layers[1].parameters()).data_ptr() == \
+        next(mono.layers[1].parameters()).data_ptr()
