@@ -35,7 +35,14 @@ class PipelineModule(nn.Module):
         self.activation_checkpoint_interval = activation_checkpoint_interval
 
         if partition_method == "parameters":
-            weights = [max(1, s.param_count()) if isinstance(s, LayerSpec) else 1 for s in self.specs]
+            def _w(s):
+                if isinstance(s, LayerSpec):
+                    return max(1, s.param_count())
+                if isinstance(s, nn.Module):  # family-A raw modules
+                    return max(1, sum(p.numel() for p in s.parameters()))
+                return 1
+
+            weights = [_w(s) for s in self.specs]
             self.bounds = partition_balanced(weights, grid.num_stages)
         elif partition_method == "uniform":
             self.bounds = partition_uniform(len(self.specs), grid.num_stages)
