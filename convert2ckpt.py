@@ -53,8 +53,14 @@ def iter_hf_state_dict(hf_dir: Path):
 
 
 def write_ckpt(outdir: Path, num_layers: int, collect, pad_vocab_to: int = 0,
-               dtype: torch.dtype | None = None, tag: str = "global_step001") -> None:
-    """collect: dict mapping flat-layer-index -> state dict (built by caller)."""
+               dtype: torch.dtype | None = None, tag: str = "global_step001",
+               mp_world_size: int = 1) -> None:
+    """collect: dict mapping flat-layer-index -> state dict (built by caller).
+
+    ``mp_world_size`` writes one ``mp_rank_{r:02d}_model_states.pt`` metadata
+    file per (vestigial) model-parallel rank, as the reference converter does
+    (convert2ckpt.py:16,38-48) — no tensor-parallel compute exists in either
+    framework; the extra files are loader-compat metadata only."""
     step_dir = outdir / tag
     step_dir.mkdir(parents=True, exist_ok=True)
     for idx, sd in collect.items():
@@ -67,14 +73,15 @@ def write_ckpt(outdir: Path, num_layers: int, collect, pad_vocab_to: int = 0,
         torch.save(sd, step_dir / f"layer_{idx:02d}-model_00-model_states.pt")
     meta = {
         "dp_world_size": 1,
-        "mp_world_size": 1,
+        "mp_world_size": mp_world_size,
         "module": None,
         "optimizer": None,
         "global_steps": 1,
         "skipped_steps": 1,
         "iteration": 1,
     }
-    torch.save(meta, step_dir / "mp_rank_00_model_states.pt")
+    for r in range(max(mp_world_size, 1)):
+        torch.save(meta, step_dir / f"mp_rank_{r:02d}_model_states.pt")
     (outdir / "latest").write_text(tag)
 
 
@@ -99,7 +106,8 @@ def copy_tokenizer_and_config(hf_dir: Path, outdir: Path) -> None:
             shutil.copy2(f, outdir / name)
 
 
-def convert_hf(hf_dir: Path, outdir: Path, pad_vocab_to: int, dtype) -> None:
+def convert_hf(hf_dir: Path, outdir: Path, pad_vocab_to: int, dtype,
+               mp_world_size: int = 1) -> None:
     layer_re = re.compile(r"^model\.layers\.(\d+)\.(.+)$")
     collect: dict[int, dict] = {}
     num_layers = 0
@@ -121,12 +129,14 @@ def convert_hf(hf_dir: Path, outdir: Path, pad_vocab_to: int, dtype) -> None:
     # renumber norm/head now that L is known
     collect[num_layers + 1] = collect.pop("norm")
     collect[num_layers + 2] = collect.pop("head")
-    write_ckpt(outdir, num_layers, collect, pad_vocab_to, dtype)
+    write_ckpt(outdir, num_layers, collect, pad_vocab_to, dtype,
+               mp_world_size=mp_world_size)
     copy_tokenizer_and_config(hf_dir, outdir)
     print(f"wrote {num_layers + 3} layer files to {outdir}")
 
 
-def convert_random(model_name: str, outdir: Path, dtype, seed: int = 0) -> None:
+def convert_random(model_name: str, outdir: Path, dtype, seed: int = 0,
+                   mp_world_size: int = 1) -> None:
     from lpp_amd.config import model_config
     from lpp_amd.layer_spec import LayerSpec
     from lpp_amd.models import deterministic_layer_init, get_layers_from_config
@@ -138,7 +148,8 @@ def convert_random(model_name: str, outdir: Path, dtype, seed: int = 0) -> None:
         layer = spec.build()
         deterministic_layer_init(layer, cfg, seed, idx)
         collect[idx] = {k: v.clone() for k, v in layer.state_dict().items()}
-    write_ckpt(outdir, cfg.num_layers, collect, 0, dtype)
+    write_ckpt(outdir, cfg.num_layers, collect, 0, dtype,
+               mp_world_size=mp_world_size)
     print(f"wrote random-init {model_name} ({cfg.num_layers + 3} layer files) to {outdir}")
 
 
@@ -151,6 +162,9 @@ def main() -> int:
     ap.add_argument("--pad-vocab-to", type=int, default=0)
     ap.add_argument("--dtype", type=str, default=None, choices=[None, "fp16", "bf16", "fp32"])
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--mp_world_size", type=int, default=1,
+                    help="emit one mp_rank_XX metadata file per rank "
+                         "(reference convert2ckpt.py:16; metadata only)")
     args = ap.parse_args()
 
     dt = {None: None, "fp16": torch.float16, "bf16": torch.bfloat16, "fp32": torch.float32}[
@@ -158,9 +172,11 @@ def main() -> int:
     ]
     out = Path(args.output_dir)
     if args.hf_dir:
-        convert_hf(Path(args.hf_dir), out, args.pad_vocab_to, dt)
+        convert_hf(Path(args.hf_dir), out, args.pad_vocab_to, dt,
+                   mp_world_size=args.mp_world_size)
     else:
-        convert_random(args.random_init, out, dt, args.seed)
+        convert_random(args.random_init, out, dt, args.seed,
+                       mp_world_size=args.mp_world_size)
     return 0
 
 

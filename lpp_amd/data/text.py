@@ -186,13 +186,25 @@ class PlaceholderDataset(Dataset):
 
 
 def completion_labels(input_ids: torch.Tensor, prompt_lens: torch.Tensor,
-                      pad_token_id: int, ignore_index: int = IGNORE_INDEX) -> torch.Tensor:
-    """labels = input_ids with pad positions AND positions < prompt_len
-    masked (loss on completion tokens only; reference get_lm_labels,
-    data/flan.py:181-190)."""
+                      pad_token_id: int, ignore_index: int = IGNORE_INDEX,
+                      lengths: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """labels = input_ids with padding AND positions < prompt_len masked
+    (loss on completion tokens only; reference get_lm_labels,
+    data/flan.py:181-190).
+
+    Padding is masked BY POSITION when ``lengths`` (real tokens per row) is
+    given: position >= length is padding.  Masking by token id alone breaks
+    when pad_token falls back to eos_token (expand_special_tokenizer) — the
+    EOS terminating each example would be masked and the model could never
+    learn to stop.  Without ``lengths`` the id-based mask is kept for
+    callers whose pad id is distinct."""
     labels = input_ids.clone()
-    keep = labels.ne(pad_token_id)
-    keep &= torch.arange(labels.size(1))[None, :] >= prompt_lens[:, None]
+    pos = torch.arange(labels.size(1), device=labels.device)[None, :]
+    if lengths is not None:
+        keep = pos < lengths.to(labels.device)[:, None]
+    else:
+        keep = labels.ne(pad_token_id)
+    keep &= pos >= prompt_lens.to(labels.device)[:, None]
     return labels.masked_fill(~keep, ignore_index).contiguous()
 
 
@@ -216,11 +228,32 @@ class Seq2SeqToCausalLM:
                                  max_length=self.max_seq_length, padding="longest",
                                  truncation=True, return_tensors="pt",
                                  add_special_tokens=True)
-        prompt_lens = prompts["input_ids"].ne(self.tokenizer.pad_token_id).sum(dim=1)
+        ids = enc["input_ids"]
+        # Real row lengths by position, not by pad-id (pad may alias eos).
+        if "attention_mask" in enc:
+            lengths = enc["attention_mask"].sum(dim=1)
+        else:
+            lengths = ids.ne(self.tokenizer.pad_token_id).sum(dim=1)
+        if "attention_mask" in prompts:
+            standalone_lens = prompts["attention_mask"].sum(dim=1)
+        else:
+            standalone_lens = prompts["input_ids"].ne(
+                self.tokenizer.pad_token_id).sum(dim=1)
+        # BPE/SentencePiece can merge tokens across the prompt/completion
+        # boundary, so the standalone prompt encoding need not be a prefix of
+        # the concatenated encoding.  Use the longest common prefix between
+        # the two encodings as the loss boundary: a boundary-merged token
+        # counts as completion (conservative — loss starts at the merge).
+        prompt_lens = torch.empty_like(standalone_lens)
+        pids = prompts["input_ids"]
+        for i in range(ids.size(0)):
+            n = int(min(standalone_lens[i], lengths[i]))
+            eq = ids[i, :n].eq(pids[i, :n])
+            prompt_lens[i] = n if bool(eq.all()) else int(eq.logical_not().byte().argmax())
         prompt_lens = torch.minimum(
             prompt_lens, torch.full_like(prompt_lens, self.max_seq_length)
         )
-        return {"input_ids": enc["input_ids"], "prompt_lens": prompt_lens}
+        return {"input_ids": ids, "prompt_lens": prompt_lens, "lengths": lengths}
 
 
 class TextCollator:
@@ -246,7 +279,8 @@ class TextCollator:
             pad = torch.full((ids.size(0), self.max_seq_length - ids.size(1)),
                              tok.pad_token_id, dtype=ids.dtype)
             ids = torch.cat([ids, pad], dim=1)
-        labels = completion_labels(ids, enc["prompt_lens"], tok.pad_token_id)
+        labels = completion_labels(ids, enc["prompt_lens"], tok.pad_token_id,
+                                   lengths=enc["lengths"])
         return {"input_ids": ids, "labels": labels}
 
 
@@ -301,6 +335,8 @@ class SimpleTokenizer:
         L = max(len(r) for r in rows)
         pad_id = self.pad_token_id if self.pad_token_id is not None else 0
         out = torch.full((len(rows), L), pad_id, dtype=torch.long)
+        mask = torch.zeros((len(rows), L), dtype=torch.long)
         for i, r in enumerate(rows):
             out[i, : len(r)] = torch.tensor(r, dtype=torch.long)
-        return {"input_ids": out}
+            mask[i, : len(r)] = 1
+        return {"input_ids": out, "attention_mask": mask}

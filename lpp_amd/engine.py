@@ -94,6 +94,13 @@ class _Watchdog:
         with self._cv:
             self._deadline = None
 
+    def stop(self):
+        with self._cv:
+            self._stop = True
+            self._deadline = None
+            self._cv.notify()
+        self._thread.join(timeout=2.0)
+
     def _run(self):
         while True:
             with self._cv:
@@ -166,7 +173,13 @@ class PipelineEngine:
         self.timers = {k: 0.0 for k in ("forward", "backward", "p2p", "allreduce", "optimizer")}
         self.schedule_position = "idle"
         self._pending_dbg = ()
-        wd_s = float(os.environ.get("LPP_WATCHDOG_S", getattr(config, "watchdog_timeout_s", 0) or 0))
+        try:
+            wd_s = float(os.environ.get("LPP_WATCHDOG_S", "") or
+                         getattr(config, "watchdog_timeout_s", 0) or 0)
+        except ValueError:
+            logger.warning("unparseable LPP_WATCHDOG_S=%r; watchdog from config",
+                           os.environ.get("LPP_WATCHDOG_S"))
+            wd_s = float(getattr(config, "watchdog_timeout_s", 0) or 0)
         self.watchdog = _Watchdog(self, wd_s) if wd_s > 0 else None
 
     # ------------------------------------------------------------------

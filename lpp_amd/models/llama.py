@@ -226,6 +226,15 @@ def get_layers_from_config(
     (no recompute, ~+25% throughput); checkpoint only the layers the memory
     budget demands (the reference's all-or-nothing flag was an 80 GB-HBM
     coping mechanism, conf/...yaml:19)."""
+    if cfg.tie_word_embeddings:
+        # Tying embedding and LM head across pipeline stages would need a
+        # tied-weight all-reduce the reference deliberately avoids for LLaMA
+        # (README.md:44-46: never tie for LLaMA) — fail loudly instead of
+        # silently training untied.
+        raise ValueError(
+            "tie_word_embeddings=True is not supported: LLaMA checkpoints are "
+            "untied and the reference (README.md:44-46) warns against tying "
+            "under pipeline parallelism. Set it to False.")
     specs: List[LayerSpec] = [LayerSpec(EmbeddingPipe, cfg.vocab_size, cfg.hidden_size)]
     for i in range(cfg.num_layers):
         ck = checkpoint_fn(i) if checkpoint_fn is not None else activation_checkpointing

@@ -70,6 +70,15 @@ class PipelineModule(nn.Module):
         self.layers = nn.ModuleList(built)
         if device is not None or dtype is not None:
             self.layers.to(device=device, dtype=dtype)
+        # Family-A (raw-module specs): drop references to OFF-STAGE modules so
+        # the monolithic source model's other-stage weights become collectable
+        # once the caller releases its own handle.  The reference materialises
+        # the full model per worker and keeps it (README.md:21); we keep only
+        # this stage's slice.  LayerSpec entries are weightless and stay.
+        self.num_specs = len(self.specs)
+        for i, s in enumerate(self.specs):
+            if isinstance(s, nn.Module) and not (self.local_start <= i < self.local_stop):
+                self.specs[i] = None
 
     # -- queries -----------------------------------------------------------
     @property
@@ -120,5 +129,5 @@ class PipelineModule(nn.Module):
     def extra_repr(self) -> str:
         return (
             f"stage={self.grid.stage_id}/{self.grid.num_stages} "
-            f"layers=[{self.local_start},{self.local_stop}) of {len(self.specs)}"
+            f"layers=[{self.local_start},{self.local_stop}) of {self.num_specs}"
         )

@@ -89,6 +89,7 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
         if done:
             break
         it = None
+        eval_it = None
         if needs_data:
             loader = build_loader(
                 dataset, cfg.micro_batch_size, grid.dp_degree, grid.dp_id,
@@ -96,6 +97,16 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
                 collator=CausalLMCollator(cfg.seq_len), epoch=epoch,
             )
             it = iter(RepeatingLoader(loader))
+            if cfg.eval_steps:
+                # A SEPARATE iterator for eval: drawing eval microbatches from
+                # the training iterator would desynchronize the resume
+                # fast-forward (which drains exactly gas draws per step).
+                eval_loader = build_loader(
+                    dataset, cfg.micro_batch_size, grid.dp_degree, grid.dp_id,
+                    seed=cfg.seed + 7919, num_workers=0,
+                    collator=CausalLMCollator(cfg.seq_len), epoch=epoch,
+                )
+                eval_it = iter(RepeatingLoader(eval_loader))
         for _ in range(steps_per_epoch):
             if step >= total_steps:
                 done = True
@@ -111,8 +122,11 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
             loss = engine.train_batch(it)
             step += 1
             tr_loss += float(loss)
-            if cfg.eval_steps and step % cfg.eval_steps == 0 and it is not None:
-                ev = float(engine.eval_batch(it, cfg.eval_micro_batches))
+            if cfg.eval_steps and step % cfg.eval_steps == 0:
+                # eval_batch is collective across the pipe group (p2p +
+                # loss broadcast): EVERY rank must enter it.  Middle stages
+                # pass eval_it=None — they never touch the iterator.
+                ev = float(engine.eval_batch(eval_it, cfg.eval_micro_batches))
                 if rank0:
                     logger.info("eval @ step %d: loss %.4f", step, ev)
                     if wandb:

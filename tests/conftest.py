@@ -10,6 +10,7 @@ if str(REPO_ROOT) not in sys.path:
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU (run on the GPU box)")
+    config.addinivalue_line("markers", "slow: long-running CPU test")
 
 
 def pytest_collection_modifyitems(config, items):
