@@ -76,6 +76,15 @@ def main() -> int:
     ap.add_argument("--act-ckpt", type=str, default="auto", choices=("auto", "0", "1"),
                     help="activation checkpointing: auto = only when the full "
                          "activations would not fit in 288 GB HBM")
+    ap.add_argument("--p2p-overlap", type=int, default=1,
+                    help="1 = pre-posted p2p on dedicated channels (default), "
+                         "0 = serial blocking exchanges (A/B baseline)")
+    ap.add_argument("--overlap-allreduce", type=int, default=1,
+                    help="1 = DP bucket all-reduce launched during the final "
+                         "backward (default), 0 = boundary all-reduce")
+    ap.add_argument("--watchdog", type=float, default=900.0,
+                    help="seconds before the deadlock watchdog dumps per-rank "
+                         "schedule position (0 = off)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -124,6 +133,10 @@ def main() -> int:
         seq_len=args.seq_len,
         dtype=args.dtype if on_gpu else "fp32",
         activation_checkpoint_interval=0,  # per-layer selective instead
+        p2p_overlap=bool(args.p2p_overlap),
+        overlap_allreduce=bool(args.overlap_allreduce),
+        watchdog_timeout_s=args.watchdog,  # armed by default: the first
+        # multi-GPU deadlock dumps per-rank schedule positions
     )
     cfg.optimizer.lr = 1e-5
     cfg.optimizer.total_num_steps = 1000
@@ -183,8 +196,10 @@ def main() -> int:
     tokens_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
-    # per-rank section timers (host-side, seconds over the timed steps) so a
-    # multi-GPU run's bubble/p2p/compute split is diagnosable from the JSON
+    # per-rank section timers (DEVICE-measured hipEvent pairs, seconds over
+    # the timed steps) so a multi-GPU run's bubble/p2p/compute split is
+    # diagnosable from the JSON: fwd/bwd are compute-stream kernel time,
+    # p2p/allreduce are comm-wait stalls suffered by the compute stream
     tsum = engine.timer_summary()
     tvec = torch.tensor([tsum["forward"], tsum["backward"], tsum["p2p"],
                          tsum["allreduce"], tsum["optimizer"]], dtype=torch.float64)
@@ -217,6 +232,13 @@ def main() -> int:
             "data": "synthetic",
             "config": {
                 "model": args.model,
+                # honest shape label: below 8 GPUs this measures the first
+                # num_layers-layer slice of the 65B model (weak scaling per
+                # stage); only n_gpus=8/pp8 runs the full headline model
+                "measured_model": (
+                    args.model if num_layers >= model_config(args.model).num_layers
+                    else f"{args.model}-shaped {num_layers}-layer slice "
+                         f"({num_layers}/{model_config(args.model).num_layers} layers)"),
                 "global_batch": args.gas * args.micro_batch_size * dp,
                 "seq_len": args.seq_len,
                 "parallelism": par,

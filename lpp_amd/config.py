@@ -177,7 +177,14 @@ class TrainConfig:
 
     # Comm
     backend: str = "nccl"  # RCCL on ROCm; "gloo" for CPU tests
+    # Overlap inter-stage p2p with compute: receives are pre-posted one
+    # microbatch ahead on dedicated fwd/bwd RCCL channels and waited at the
+    # point of use; sends are drained at the step boundary (engine.py).
     p2p_overlap: bool = True
+    # Launch DP gradient-bucket all-reduces as the FINAL backward retires
+    # each bucket (reverse-layer order) instead of all at the boundary —
+    # the reference's overlap_comm: True (conf/...yaml:154-159).
+    overlap_allreduce: bool = True
     eval_steps: int = 0  # run a forward-only eval pass every N steps
     eval_micro_batches: int = 8
     checkpoint_layers_per_stage: int = -1  # -1 = activation_checkpoint_interval rules

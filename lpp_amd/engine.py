@@ -37,6 +37,7 @@ from .optim import MixedPrecisionAdamW, WarmupDecayLR
 from .p2p import PipeP2P
 from .pipeline_module import PipelineModule
 from .topology import ProcessGrid
+from .utils.timers import DeviceTimers
 
 logger = logging.getLogger(__name__)
 
@@ -169,10 +170,42 @@ class PipelineEngine:
         self.global_steps = 0
         self.skipped_steps = 0
         self._step_time = 0.0
-        # observability
-        self.timers = {k: 0.0 for k in ("forward", "backward", "p2p", "allreduce", "optimizer")}
+        # observability — device-true section timers
+        # (hipEvent pairs on GPU; round-1's host timers bracketed enqueue
+        # only and misattributed all async work to "backward")
+        self.device_timers = DeviceTimers(device)
         self.schedule_position = "idle"
         self._pending_dbg = ()
+
+        # ---- DP gradient-bucket overlap plan (reference overlap_comm: True,
+        # conf/...yaml:154-159).  Buckets are built over the contiguous flat
+        # fp32 grad buffer in REVERSE parameter order — the order the final
+        # backward retires them — so each bucket's all-reduce can launch as
+        # soon as its last gradient lands, overlapping the remaining
+        # backward compute on the comm stream.
+        self._dp_overlap = bool(
+            config.overlap_allreduce and grid.dp_degree > 1 and not self.zero1
+        )
+        self._buckets = []           # dicts: start,end,count,remaining,launched
+        self._bucket_of = {}         # id(param) -> bucket index
+        self._ar_handles = []
+        self._grad_finalizing = False
+        if self._dp_overlap:
+            bucket_elems = max(1, config.allreduce_bucket_mb * 1024 * 1024 // 4)
+            offs, off = {}, 0
+            for p in self.optimizer.params:
+                offs[id(p)] = off
+                off += p.numel()
+            cur, cur_elems = [], 0
+            for p in reversed(self.optimizer.params):
+                cur.append(p)
+                cur_elems += p.numel()
+                if cur_elems >= bucket_elems:
+                    self._push_bucket(cur, offs)
+                    cur, cur_elems = [], 0
+            if cur:
+                self._push_bucket(cur, offs)
+            self.optimizer.on_accumulate = self._on_grad_accumulated
         try:
             wd_s = float(os.environ.get("LPP_WATCHDOG_S", "") or
                          getattr(config, "watchdog_timeout_s", 0) or 0)
@@ -183,6 +216,40 @@ class PipelineEngine:
         self.watchdog = _Watchdog(self, wd_s) if wd_s > 0 else None
 
     # ------------------------------------------------------------------
+    def _push_bucket(self, params, offs) -> None:
+        idx = len(self._buckets)
+        start = min(offs[id(p)] for p in params)
+        end = max(offs[id(p)] + p.numel() for p in params)
+        self._buckets.append(
+            {"start": start, "end": end, "count": len(params),
+             "remaining": len(params), "launched": False}
+        )
+        for p in params:
+            self._bucket_of[id(p)] = idx
+
+    def _on_grad_accumulated(self, p) -> None:
+        """Optimizer post-accumulate observer: during the FINAL microbatch's
+        backward, launch a bucket's DP all-reduce the moment its last
+        gradient is final (runs on the autograd thread, async_op — the
+        collective rides the comm stream while backward keeps computing)."""
+        if not self._grad_finalizing:
+            return
+        b = self._buckets[self._bucket_of[id(p)]]
+        b["remaining"] -= 1
+        if b["remaining"] == 0 and not b["launched"]:
+            b["launched"] = True
+            chunk = self.optimizer.flat_grads.narrow(0, b["start"], b["end"] - b["start"])
+            self._ar_handles.append(
+                dist.all_reduce(chunk, op=dist.ReduceOp.SUM, group=self.grid.dp_group,
+                                async_op=True)
+            )
+
+    @property
+    def timers(self) -> dict:
+        base = {k: 0.0 for k in ("forward", "backward", "p2p", "allreduce", "optimizer")}
+        base.update(self.device_timers.totals_nosync())
+        return base
+
     @property
     def is_first_stage(self) -> bool:
         return self.grid.is_first_stage()
@@ -242,74 +309,133 @@ class PipelineEngine:
 
     # -- the 1F1B schedule --------------------------------------------------
     def train_batch(self, data_iter: Iterator) -> torch.Tensor:
-        """One optimizer step == ``micro_batches`` microbatches, 1F1B."""
+        """One optimizer step == ``micro_batches`` microbatches, 1F1B.
+
+        With ``p2p_overlap`` (default) every network receive is PRE-POSTED
+        one use ahead on its dedicated channel (PipeP2P docstring) and
+        waited only at the point of use, so the xGMI transfer of the next
+        microbatch's activation rides under this microbatch's backward, and
+        sends are fire-and-forget (drained at the step boundary).  With the
+        flag off, each receive is posted-and-waited at its use site and
+        sends are waited immediately — the strictly serial reference
+        ordering, kept as the A/B baseline."""
         t0 = time.time()
         if self.watchdog:
             self.watchdog.arm()
         self.module.train()
+        timers = self.device_timers
         M = self.micro_batches
         P = self.grid.num_stages
         s = self.grid.stage_id
         warmup = min(P - 1 - s, M)
         remaining = M - warmup
+        overlap = bool(self.config.p2p_overlap)
 
         pending = deque()  # (input_tensor, backward_handle)
         losses = []
+        sends = []  # in-flight sends (overlap mode)
+
+        first, last = self.is_first_stage, self.is_last_stage
+        acts_needed = 0 if first else M     # network activation recvs
+        grads_needed = 0 if last else M     # network gradient recvs
+        acts_posted = grads_posted = 0
+        pend_act = pend_grad = None
+        backwards_done = 0
+
+        def post_act():
+            nonlocal acts_posted
+            acts_posted += 1
+            return self.p2p.irecv_forward()
+
+        def post_grad():
+            nonlocal grads_posted
+            grads_posted += 1
+            return self.p2p.irecv_backward()
+
+        def do_send(pending_send):
+            # Sends are fire-and-forget in BOTH modes (drained at the step
+            # boundary): waiting a send in-line would rendezvous against the
+            # peer's recv-post order and can deadlock on gloo.  The overlap
+            # flag gates only the pre-posting of receives.
+            if pending_send is not None:
+                sends.append(pending_send)
+
+        if overlap and acts_posted < acts_needed:
+            pend_act = post_act()
+
+        def take_act():
+            nonlocal pend_act
+            if first:
+                return None
+            with timers.section("p2p"):
+                p = pend_act if pend_act is not None else post_act()
+                x = p.wait()
+            pend_act = post_act() if (overlap and acts_posted < acts_needed) else None
+            return x
+
+        def take_grad():
+            nonlocal pend_grad
+            if last:
+                return None
+            with timers.section("p2p"):
+                p = pend_grad if pend_grad is not None else post_grad()
+                g = p.wait()
+            pend_grad = post_grad() if (overlap and grads_posted < grads_needed) else None
+            return g
+
+        def run_backward(recv_grad):
+            nonlocal backwards_done
+            b_inp, b_handle = pending.popleft()
+            if backwards_done + 1 == M and self._dp_overlap and dist.is_initialized():
+                self._grad_finalizing = True
+            with timers.section("backward"):
+                g = self._backward_step(b_inp, b_handle, recv_grad)
+            backwards_done += 1
+            self._grad_finalizing = False
+            if g is not None:
+                with timers.section("p2p"):
+                    do_send(self.p2p.isend_backward(g))
 
         # ---- warmup forwards
         for wi in range(warmup):
             self.schedule_position = f"warmup fwd {wi + 1}/{warmup}"
-            tp = time.time()
-            x = self.p2p.recv_forward()
-            self.timers["p2p"] += time.time() - tp
-            tf = time.time()
-            inp, handle, loss = self._forward_step(x, data_iter)
-            self.timers["forward"] += time.time() - tf
-            tp = time.time()
-            self.p2p.send_forward(handle if not self.is_last_stage else None)
-            self.timers["p2p"] += time.time() - tp
+            x = take_act()
+            with timers.section("forward"):
+                inp, handle, loss = self._forward_step(x, data_iter)
+            if loss is not None:
+                losses.append(loss)
+            if not last:
+                with timers.section("p2p"):
+                    do_send(self.p2p.isend_forward(handle))
             pending.append((inp, handle))
             self._pending_dbg = tuple(range(len(pending)))
 
-        x = self.p2p.recv_forward() if remaining > 0 else None
+        if overlap and grads_posted < grads_needed:
+            pend_grad = post_grad()
 
         # ---- steady 1F1B
         for i in range(remaining):
             self.schedule_position = f"steady 1F1B {i + 1}/{remaining}"
-            tf = time.time()
-            inp, handle, loss = self._forward_step(x, data_iter)
-            self.timers["forward"] += time.time() - tf
-            pending.append((inp, handle))
+            x = take_act()
+            with timers.section("forward"):
+                inp, handle, loss = self._forward_step(x, data_iter)
             if loss is not None:
                 losses.append(loss)
-            if self.is_last_stage:
-                recv_grad = None
-            else:
-                recv_grad = self.p2p.send_forward_recv_backward(handle)
-            b_inp, b_handle = pending.popleft()
-            tb = time.time()
-            g = self._backward_step(b_inp, b_handle, recv_grad)
-            self.timers["backward"] += time.time() - tb
-            last = i == remaining - 1
-            if g is None:  # first stage sends nothing backward
-                x = None if last else self.p2p.recv_forward()
-            elif last:
-                self.p2p.send_backward(g)
-            else:
-                x = self.p2p.send_backward_recv_forward(g)
+            if not last:
+                with timers.section("p2p"):
+                    do_send(self.p2p.isend_forward(handle))
+            pending.append((inp, handle))
+            run_backward(take_grad())
 
         # ---- cooldown backwards
         for ci in range(warmup):
             self.schedule_position = f"cooldown bwd {ci + 1}/{warmup}"
-            b_inp, b_handle = pending.popleft()
-            tp = time.time()
-            recv_grad = self.p2p.recv_backward() if not self.is_last_stage else None
-            self.timers["p2p"] += time.time() - tp
-            tb = time.time()
-            g = self._backward_step(b_inp, b_handle, recv_grad)
-            self.timers["backward"] += time.time() - tb
-            if g is not None:
-                self.p2p.send_backward(g)
+            run_backward(take_grad())
+
+        # drain outstanding sends before touching the buffers' storages
+        with timers.section("p2p"):
+            for sp in sends:
+                sp.wait()
 
         # ---- boundary: DP all-reduce, clip, step
         self.schedule_position = "optimizer"
@@ -379,55 +505,71 @@ class PipelineEngine:
                 dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=self.grid.dp_group)
             opt.grad_shard.div_(self.grid.dp_degree)
             return
-        bucket_elems = max(1, self.config.allreduce_bucket_mb * 1024 * 1024 // 4)
-        handles = []
-        for off in range(0, flat.numel(), bucket_elems):
-            chunk = flat.narrow(0, off, min(bucket_elems, flat.numel() - off))
-            handles.append(
-                dist.all_reduce(chunk, op=dist.ReduceOp.SUM, group=self.grid.dp_group,
-                                async_op=True)
-            )
-        for h in handles:
-            h.wait()
+        if self._dp_overlap:
+            # buckets not retired by the final backward's hooks (shouldn't
+            # happen, but a param skipped in the graph would strand one)
+            for b in self._buckets:
+                if not b["launched"]:
+                    b["launched"] = True
+                    chunk = flat.narrow(0, b["start"], b["end"] - b["start"])
+                    self._ar_handles.append(
+                        dist.all_reduce(chunk, op=dist.ReduceOp.SUM,
+                                        group=self.grid.dp_group, async_op=True)
+                    )
+            for h in self._ar_handles:
+                h.wait()
+            self._ar_handles.clear()
+            for b in self._buckets:
+                b["remaining"] = b["count"]
+                b["launched"] = False
+        else:
+            bucket_elems = max(1, self.config.allreduce_bucket_mb * 1024 * 1024 // 4)
+            handles = []
+            for off in range(0, flat.numel(), bucket_elems):
+                chunk = flat.narrow(0, off, min(bucket_elems, flat.numel() - off))
+                handles.append(
+                    dist.all_reduce(chunk, op=dist.ReduceOp.SUM, group=self.grid.dp_group,
+                                    async_op=True)
+                )
+            for h in handles:
+                h.wait()
         flat.div_(self.grid.dp_degree)
 
     def _optimizer_step(self) -> None:
-        ta = time.time()
-        self._allreduce_gradients()
-        self.timers["allreduce"] += time.time() - ta
-        t_o = time.time()
+        timers = self.device_timers
+        with timers.section("allreduce"):
+            self._allreduce_gradients()
+        with timers.section("optimizer"):
+            inv_scale = 1.0
+            if self.loss_scaler is not None:
+                inv_scale = 1.0 / self.loss_scaler.scale
 
-        inv_scale = 1.0
-        if self.loss_scaler is not None:
-            inv_scale = 1.0 / self.loss_scaler.scale
+            sq = self.optimizer.grad_sq_sum()
+            if dist.is_initialized():
+                if self.zero1:
+                    # shards tile (dp x stage): sum across the whole world
+                    dist.all_reduce(sq, op=dist.ReduceOp.SUM)
+                elif self.grid.num_stages > 1:
+                    dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.grid.pipe_group)
+            global_norm = (sq.float().sqrt() * inv_scale).item()
 
-        sq = self.optimizer.grad_sq_sum()
-        if dist.is_initialized():
-            if self.zero1:
-                # shards tile (dp x stage): sum across the whole world
-                dist.all_reduce(sq, op=dist.ReduceOp.SUM)
-            elif self.grid.num_stages > 1:
-                dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.grid.pipe_group)
-        global_norm = (sq.float().sqrt() * inv_scale).item()
+            if self.loss_scaler is not None:
+                found_inf = not (global_norm == global_norm and global_norm != float("inf"))
+                self.loss_scaler.update(found_inf)
+                if found_inf:
+                    self.optimizer.zero_grad()
+                    self.skipped_steps += 1
+                    self.global_steps += 1
+                    return
 
-        if self.loss_scaler is not None:
-            found_inf = not (global_norm == global_norm and global_norm != float("inf"))
-            self.loss_scaler.update(found_inf)
-            if found_inf:
-                self.optimizer.zero_grad()
-                self.skipped_steps += 1
-                self.global_steps += 1
-                return
-
-        clip = self.config.optimizer.max_grad_norm
-        coef = inv_scale
-        if clip > 0 and global_norm > clip:
-            coef *= clip / (global_norm + 1e-6)
-        self.optimizer.step(grad_scale=coef)
-        self.optimizer.zero_grad()
-        self.lr_scheduler.step()
-        self.global_steps += 1
-        self.timers["optimizer"] += time.time() - t_o
+            clip = self.config.optimizer.max_grad_norm
+            coef = inv_scale
+            if clip > 0 and global_norm > clip:
+                coef *= clip / (global_norm + 1e-6)
+            self.optimizer.step(grad_scale=coef)
+            self.optimizer.zero_grad()
+            self.lr_scheduler.step()
+            self.global_steps += 1
 
     # ------------------------------------------------------------------
     @property
@@ -435,14 +577,14 @@ class PipelineEngine:
         return self._step_time
 
     def timer_summary(self, reset: bool = True) -> dict:
-        """Host-side accumulated section times (seconds) since last reset.
-        Note these bracket ENQUEUE + sync points of async GPU work — use
-        rocprofv3 for device-side truth; these locate stalls per stage."""
-        out = dict(self.timers)
-        if reset:
-            for k in self.timers:
-                self.timers[k] = 0.0
-        return out
+        """Device-true accumulated section times (seconds) since last reset.
+        On GPU these are hipEvent pairs recorded on the compute stream —
+        forward/backward are real kernel time, p2p/allreduce are the stalls
+        the compute stream suffered waiting on the comm streams.  Folding
+        synchronizes the device."""
+        base = {k: 0.0 for k in ("forward", "backward", "p2p", "allreduce", "optimizer")}
+        base.update(self.device_timers.summary(reset=reset))
+        return base
 
     def get_lr(self) -> float:
         return self.optimizer.lr

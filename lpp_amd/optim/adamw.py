@@ -108,15 +108,20 @@ class MixedPrecisionAdamW:
                 self.exp_avg.append(torch.zeros_like(self.masters[-1]))
                 self.exp_avg_sq.append(torch.zeros_like(self.masters[-1]))
                 off += n
+        # Optional observer fired after each grad lands in main_grad — the
+        # engine uses it to launch DP bucket all-reduces as the FINAL
+        # microbatch's backward retires each parameter (overlap_allreduce).
+        self.on_accumulate = None
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._accumulate_hook) for p in self.params
         ]
 
-    @staticmethod
-    def _accumulate_hook(p: torch.nn.Parameter) -> None:
+    def _accumulate_hook(self, p: torch.nn.Parameter) -> None:
         if p.grad is not None:
             p.main_grad.add_(p.grad)
             p.grad = None
+            if self.on_accumulate is not None:
+                self.on_accumulate(p)
 
     # ------------------------------------------------------------------
     @property

@@ -9,23 +9,51 @@ Native replacement for the p2p DeepSpeed's PipelineEngine performs inside
 - Static shapes: the engine knows (mbs, S, H) up front, so there is no
   per-tensor meta handshake on the wire; recv buffers come from the caching
   allocator.
-- Bidirectional exchanges (send fwd + recv bwd at the 1F1B steady state) are
-  posted as one ``batch_isend_irecv`` group, which maps to a single
-  ncclGroup on RCCL — both directions ride the same xGMI link concurrently
-  (links are full duplex ~153 GB/s each way).
+- TWO dedicated communicators (topology.pipe_fwd_group / pipe_bwd_group):
+  activation traffic and gradient traffic ride separate RCCL comms, i.e.
+  separate internal streams.  That makes pre-posted receives safe — a recv
+  for the NEXT microbatch's activation enqueued on the fwd channel can
+  never serialize behind this microbatch's gradient send on the bwd
+  channel, which is the stream-ordering hazard that deadlocks an overlapped
+  1F1B schedule on a single comm stream.
+- Asynchronous primitives (``irecv_forward`` etc.) return a ``Pending``
+  handle; the engine posts receives one microbatch ahead and waits only at
+  the point of use, overlapping the xGMI transfer with compute
+  (SURVEY.md hard-part #2; the reference delegates this to DeepSpeed's
+  double-buffered p2p behind trainer_base_ds_mp.py:354).
 
 On a single fully-connected xGMI node every stage pair is one hop; the
-neighbour-only PP pattern gives each stage boundary a dedicated link.
+neighbour-only PP pattern gives each stage boundary a dedicated link
+(~153 GB/s each way, full duplex).
 """
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist
 
 from .topology import ProcessGrid
+
+
+class Pending:
+    """An in-flight p2p operation: ``wait()`` blocks (stream-orders, on
+    RCCL) until complete and returns the received buffer (None for sends).
+    Holds a reference to the tensor so the caching allocator cannot reuse
+    the storage while the transfer is in flight."""
+
+    __slots__ = ("req", "buf")
+
+    def __init__(self, req, buf: Optional[torch.Tensor]):
+        self.req = req
+        self.buf = buf
+
+    def wait(self) -> Optional[torch.Tensor]:
+        if self.req is not None:
+            self.req.wait()
+            self.req = None
+        return self.buf
 
 
 class PipeP2P:
@@ -35,63 +63,79 @@ class PipeP2P:
         self.act_shape = tuple(act_shape)
         self.dtype = dtype
         self.device = device
+        # channel groups (None outside an initialized process group, e.g.
+        # single-process tests — p2p is a no-op there anyway)
+        self._fwd_group = grid.pipe_fwd_group
+        self._bwd_group = grid.pipe_bwd_group
 
     def _empty(self) -> torch.Tensor:
         return torch.empty(self.act_shape, dtype=self.dtype, device=self.device)
 
-    def _run(self, ops: List[dist.P2POp]) -> None:
-        if not ops:
-            return
-        reqs = dist.batch_isend_irecv(ops)
-        for r in reqs:
-            r.wait()
-
-    # -- single-direction --------------------------------------------------
-    def send_forward(self, tensor: torch.Tensor) -> None:
+    # -- async primitives (the engine's hot path) ---------------------------
+    def isend_forward(self, tensor: torch.Tensor) -> Optional[Pending]:
+        """Post the activation send to the next stage on the fwd channel."""
         if self.grid.next_rank is None:
+            return None
+        t = tensor.contiguous()
+        req = dist.isend(t, self.grid.next_rank, group=self._fwd_group)
+        return Pending(req, t)
+
+    def irecv_forward(self) -> Optional[Pending]:
+        if self.grid.prev_rank is None:
+            return None
+        buf = self._empty()
+        req = dist.irecv(buf, self.grid.prev_rank, group=self._fwd_group)
+        return Pending(req, buf)
+
+    def isend_backward(self, grad: torch.Tensor) -> Optional[Pending]:
+        """Post the gradient send to the previous stage on the bwd channel."""
+        if self.grid.prev_rank is None:
+            return None
+        t = grad.contiguous()
+        req = dist.isend(t, self.grid.prev_rank, group=self._bwd_group)
+        return Pending(req, t)
+
+    def irecv_backward(self) -> Optional[Pending]:
+        if self.grid.next_rank is None:
+            return None
+        buf = self._empty()
+        req = dist.irecv(buf, self.grid.next_rank, group=self._bwd_group)
+        return Pending(req, buf)
+
+    # -- blocking convenience (eval/generation; not the training hot path) --
+    def send_forward(self, tensor: Optional[torch.Tensor]) -> None:
+        if tensor is None or self.grid.next_rank is None:
             return
-        self._run([dist.P2POp(dist.isend, tensor.contiguous(), self.grid.next_rank)])
+        p = self.isend_forward(tensor)
+        if p is not None:
+            p.wait()
 
     def recv_forward(self) -> Optional[torch.Tensor]:
-        if self.grid.prev_rank is None:
-            return None
-        buf = self._empty()
-        self._run([dist.P2POp(dist.irecv, buf, self.grid.prev_rank)])
-        return buf
+        p = self.irecv_forward()
+        return p.wait() if p is not None else None
 
     def send_backward(self, grad: torch.Tensor) -> None:
-        if self.grid.prev_rank is None:
-            return
-        self._run([dist.P2POp(dist.isend, grad.contiguous(), self.grid.prev_rank)])
+        p = self.isend_backward(grad)
+        if p is not None:
+            p.wait()
 
     def recv_backward(self) -> Optional[torch.Tensor]:
-        if self.grid.next_rank is None:
-            return None
-        buf = self._empty()
-        self._run([dist.P2POp(dist.irecv, buf, self.grid.next_rank)])
-        return buf
+        p = self.irecv_backward()
+        return p.wait() if p is not None else None
 
-    # -- combined (1F1B steady state) ---------------------------------------
+    # -- combined (legacy blocking 1F1B steady state) ------------------------
     def send_forward_recv_backward(self, tensor: torch.Tensor) -> Optional[torch.Tensor]:
         if self.grid.next_rank is None:
             return None
-        buf = self._empty()
-        self._run(
-            [
-                dist.P2POp(dist.isend, tensor.contiguous(), self.grid.next_rank),
-                dist.P2POp(dist.irecv, buf, self.grid.next_rank),
-            ]
-        )
-        return buf
+        s = self.isend_forward(tensor)
+        r = self.irecv_backward()
+        s.wait()
+        return r.wait()
 
     def send_backward_recv_forward(self, grad: torch.Tensor) -> Optional[torch.Tensor]:
         if self.grid.prev_rank is None:
             return None
-        buf = self._empty()
-        self._run(
-            [
-                dist.P2POp(dist.isend, grad.contiguous(), self.grid.prev_rank),
-                dist.P2POp(dist.irecv, buf, self.grid.prev_rank),
-            ]
-        )
-        return buf
+        s = self.isend_backward(grad)
+        r = self.irecv_forward()
+        s.wait()
+        return r.wait()

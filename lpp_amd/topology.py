@@ -25,6 +25,14 @@ import torch.distributed as dist
 class _Groups:
     pipe: object = None  # this rank's pipeline group (its dp column)
     data: object = None  # this rank's DP group (its stage row)
+    # Dedicated communicators for the two p2p traffic directions.  On RCCL
+    # each process group owns its own internal stream; putting activation
+    # (fwd-going) and gradient (bwd-going) traffic on separate communicators
+    # means a pre-posted activation recv can never serialize behind a
+    # gradient send on the same stream — the hazard that would deadlock an
+    # overlapped 1F1B schedule (engine.py).
+    pipe_fwd: object = None
+    pipe_bwd: object = None
 
 
 class ProcessGrid:
@@ -95,12 +103,16 @@ class ProcessGrid:
             g = dist.new_group(ranks=ranks)
             if self.rank in ranks:
                 self._groups.data = g
-        # Pipe groups: one per dp column.
+        # Pipe groups: one per dp column (plus the two p2p channel groups).
         for dp in range(self.dp_degree):
             ranks = [s * self.dp_degree + dp for s in range(self.num_stages)]
             g = dist.new_group(ranks=ranks)
+            gf = dist.new_group(ranks=ranks)
+            gb = dist.new_group(ranks=ranks)
             if self.rank in ranks:
                 self._groups.pipe = g
+                self._groups.pipe_fwd = gf
+                self._groups.pipe_bwd = gb
 
     @property
     def dp_group(self):
@@ -109,6 +121,14 @@ class ProcessGrid:
     @property
     def pipe_group(self):
         return self._groups.pipe
+
+    @property
+    def pipe_fwd_group(self):
+        return self._groups.pipe_fwd
+
+    @property
+    def pipe_bwd_group(self):
+        return self._groups.pipe_bwd
 
     def __repr__(self) -> str:
         return (

@@ -55,8 +55,11 @@ def build_engine(cfg, rank, world_size, seed=11):
     return PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
 
 
-def run_steps(rank, world_size, num_stages, steps=3, gas=4, dtype="fp32", seed=11):
+def run_steps(rank, world_size, num_stages, steps=3, gas=4, dtype="fp32", seed=11,
+              p2p_overlap=True, overlap_allreduce=True):
     cfg = make_config(num_stages=num_stages, gas=gas, dtype=dtype)
+    cfg.p2p_overlap = p2p_overlap
+    cfg.overlap_allreduce = overlap_allreduce
     engine = build_engine(cfg, rank, world_size, seed=seed)
     dp = engine.grid.dp_degree
     dp_id = engine.grid.dp_id

@@ -101,6 +101,22 @@ def test_pp4_gas_less_than_stages():
         assert abs(a - b) < 1e-3
 
 
+def test_p2p_overlap_off_matches_on():
+    """The serial (p2p_overlap=False) and overlapped schedules are the same
+    numerics — bit-identical loss trajectories."""
+    on = run_dist(2, run_steps, 2, 2, 6)
+    off = run_dist(2, run_steps, 2, 2, 6, "fp32", 11, False)
+    assert on[0] == off[0]
+
+
+def test_overlap_allreduce_off_matches_on():
+    """Bucket all-reduces launched from the final backward's hooks produce
+    the same trajectory as the boundary all-reduce (PP2 x DP2)."""
+    on = run_dist(4, run_steps, 2, 2, 4)
+    off = run_dist(4, run_steps, 2, 2, 4, "fp32", 11, True, False)
+    assert on[0] == pytest.approx(off[0], abs=1e-6)
+
+
 @pytest.mark.slow
 def test_pp8_matches_single_process():
     """world 8 = the driver's N=8 scale-run schedule shape, on gloo."""
