@@ -37,10 +37,34 @@ def _load_extension():
 
         _EXT = _build.load_extension()
         _EXT_ERR = None
+        if torch.cuda.is_available():
+            _pin_wgrad_algos(_EXT)
     except Exception as e:  # pragma: no cover - exercised on GPU box
         _EXT = None
         _EXT_ERR = f"{type(e).__name__}: {e}"
     return _EXT
+
+
+def _pin_wgrad_algos(ext) -> None:
+    """Pin committed hipBLASLt solution indices for the wgrad GEMM shapes
+    (lpp_amd/ops/wgrad_algos.json, produced by scripts/wgrad_tune.py's
+    exhaustive device-timed sweep).  Indices are library-version-specific;
+    the JSON records the hipBLASLt version it was tuned on and is skipped
+    on mismatch (falling back to first-use heuristic timing)."""
+    import json
+
+    path = os.path.join(os.path.dirname(__file__), "wgrad_algos.json")
+    if not os.path.exists(path):
+        return
+    try:
+        data = json.load(open(path))
+        for entry in data.get("shapes", []):
+            ext.wgrad_set_algo(entry["T"], entry["in"], entry["out"], entry["index"])
+    except Exception as e:  # pragma: no cover
+        import logging
+
+        logging.getLogger(__name__).warning(
+            "wgrad algo pinning skipped: %s", e)
 
 
 def force_eager() -> bool:

@@ -11,10 +11,24 @@
 // Row-major tensors mapped onto hipBLASLt's column-major view:
 //   D_rm[out,in] = dY_rm[T,out]^T * X_rm[T,in]
 //   <=> D_cm[in,out] = X_cm[in,T](opA=N) * dY_cm[out,T](opB=T)
+//
+// Algorithm selection (round 2): the wgrad GEMM classes ran 15-25% below
+// the bf16 forward/dgrad classes with the heuristic's top-48 candidates
+// (profiles/r01_65b_1gpu_step_kernel_stats_final.csv).  wgrad_tune() sweeps
+// EVERY library solution supported for the problem
+// (hipblaslt_ext::getAllAlgos + matmulIsAlgoSupported) with device-timed
+// reps; the winning solution indices are committed per shape
+// (lpp_amd/ops/wgrad_algos.json) and pinned at import via wgrad_set_algo —
+// index-based pinning is safe here because the ROCm image (hipBLASLt
+// version) is fixed.
 #include <hipblaslt/hipblaslt.h>
+#include <hipblaslt/hipblaslt-ext.hpp>
 
 #include <mutex>
+#include <string>
+#include <tuple>
 #include <unordered_map>
+#include <vector>
 
 #include "common.h"
 
@@ -55,6 +69,102 @@ struct CachedPlan {
   hipblasLtMatmulAlgo_t algo{};
 };
 
+static std::mutex g_mu;
+static std::unordered_map<AlgoKey, CachedPlan, AlgoKeyHash> g_plans;
+
+// Create desc/layouts for the wgrad problem (no algo chosen yet).
+static CachedPlan make_problem(int64_t T, int64_t in, int64_t out) {
+  CachedPlan plan;
+  LPP_CHECK_BLASLT(hipblasLtMatmulDescCreate(&plan.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  hipblasOperation_t opN = HIPBLAS_OP_N, opT = HIPBLAS_OP_T;
+  LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+      plan.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opN, sizeof(opN)));
+  LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+      plan.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opT, sizeof(opT)));
+  // cm views: A = X [in, T] ld=in; B = dY [out, T] ld=out (opB=T); C/D = [in, out] ld=in
+  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, in, T, in));
+  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, out, T, out));
+  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.c, HIP_R_32F, in, out, in));
+  return plan;
+}
+
+// Device-timed run of one candidate; returns best-of-reps milliseconds, or a
+// huge value if the algo fails at runtime.
+static float time_algo(CachedPlan& plan, hipblasLtMatmulAlgo_t& algo, void* ax, void* by,
+                       void* cw, void* ws, int reps, hipStream_t stream) {
+  const float a1 = 1.0f, b1 = 1.0f;
+  hipEvent_t ev0, ev1;
+  LPP_CHECK_HIP(hipEventCreate(&ev0));
+  LPP_CHECK_HIP(hipEventCreate(&ev1));
+  float best = 1e30f;
+  // one warmup
+  hipblasStatus_t st = hipblasLtMatmul(lt_ctx().handle, plan.op, &a1, ax, plan.a, by,
+                                       plan.b, &b1, cw, plan.c, cw, plan.c, &algo, ws,
+                                       kWorkspaceBytes, stream);
+  if (st == HIPBLAS_STATUS_SUCCESS) {
+    for (int r = 0; r < reps; ++r) {
+      LPP_CHECK_HIP(hipEventRecord(ev0, stream));
+      st = hipblasLtMatmul(lt_ctx().handle, plan.op, &a1, ax, plan.a, by, plan.b, &b1,
+                           cw, plan.c, cw, plan.c, &algo, ws, kWorkspaceBytes, stream);
+      LPP_CHECK_HIP(hipEventRecord(ev1, stream));
+      if (st != HIPBLAS_STATUS_SUCCESS) break;
+      LPP_CHECK_HIP(hipEventSynchronize(ev1));
+      float ms = 0.f;
+      LPP_CHECK_HIP(hipEventElapsedTime(&ms, ev0, ev1));
+      if (ms < best) best = ms;
+    }
+  }
+  LPP_CHECK_HIP(hipEventDestroy(ev0));
+  LPP_CHECK_HIP(hipEventDestroy(ev1));
+  return best;
+}
+
+// Default first-use selection: heuristic top-48, each device-timed once.
+static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out) {
+  hipblasLtMatmulPreference_t pref;
+  LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceCreate(&pref));
+  size_t ws = kWorkspaceBytes;
+  LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
+  hipblasLtMatmulHeuristicResult_t results[48];
+  int found = 0;
+  LPP_CHECK_BLASLT(hipblasLtMatmulAlgoGetHeuristic(
+      lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 48, results,
+      &found));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  TORCH_CHECK(found > 0, "hipblaslt: no algo for wgrad shape [", out, ",", in,
+              "] k=", T);
+  plan.algo = results[0].algo;
+  if (found > 1) {
+    auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
+    auto sx = at::empty({T, in}, opt);
+    auto sy = at::empty({T, out}, opt);
+    auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
+    auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
+    auto stream = current_stream();
+    float best = 1e30f;
+    for (int i = 0; i < found; ++i) {
+      float ms = time_algo(plan, results[i].algo, sx.data_ptr(), sy.data_ptr(),
+                           sw.data_ptr(), wsbuf.data_ptr(), 1, stream);
+      if (ms < best) {
+        best = ms;
+        plan.algo = results[i].algo;
+      }
+    }
+  }
+}
+
+static CachedPlan get_plan(int64_t T, int64_t in, int64_t out) {
+  const AlgoKey key{in, out, T};
+  std::lock_guard<std::mutex> lock(g_mu);
+  auto it = g_plans.find(key);
+  if (it != g_plans.end()) return it->second;
+  CachedPlan plan = make_problem(T, in, out);
+  pick_heuristic(plan, T, in, out);
+  g_plans.emplace(key, plan);
+  return plan;
+}
+
 // dW (row-major [out, in], fp32) += dY(row-major [T, out], bf16)^T @ X(row-major [T, in], bf16)
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.dim() == 2);
@@ -65,82 +175,7 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   TORCH_CHECK(dy.size(0) == T && dw.size(0) == out && dw.size(1) == in,
               "wgrad shapes: x[T,in] dy[T,out] dw[out,in]");
 
-  static std::mutex mu;
-  static std::unordered_map<AlgoKey, CachedPlan, AlgoKeyHash> plans;
-
-  const AlgoKey key{in, out, T};
-  CachedPlan plan;
-  {
-    std::lock_guard<std::mutex> lock(mu);
-    auto it = plans.find(key);
-    if (it != plans.end()) {
-      plan = it->second;
-    } else {
-      LPP_CHECK_BLASLT(hipblasLtMatmulDescCreate(&plan.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
-      hipblasOperation_t opN = HIPBLAS_OP_N, opT = HIPBLAS_OP_T;
-      LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
-          plan.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opN, sizeof(opN)));
-      LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
-          plan.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opT, sizeof(opT)));
-      // cm views: A = X [in, T] ld=in; B = dY [out, T] ld=out (opB=T); C/D = [in, out] ld=in
-      LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, in, T, in));
-      LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, out, T, out));
-      LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.c, HIP_R_32F, in, out, in));
-
-      hipblasLtMatmulPreference_t pref;
-      LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceCreate(&pref));
-      size_t ws = kWorkspaceBytes;
-      LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceSetAttribute(
-          pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-      hipblasLtMatmulHeuristicResult_t results[48];
-      int found = 0;
-      LPP_CHECK_BLASLT(hipblasLtMatmulAlgoGetHeuristic(
-          lt_ctx().handle, plan.op, plan.a, plan.b, plan.c, plan.c, pref, 48, results,
-          &found));
-      hipblasLtMatmulPreferenceDestroy(pref);
-      TORCH_CHECK(found > 0, "hipblaslt: no algo for wgrad shape [", out, ",", in,
-                  "] k=", T);
-      // measured pick: time each heuristic candidate once on scratch
-      // operands (first call per shape only; the heuristic's first choice
-      // was ~10% slower than the best for the 65B beta=1 f32-D shapes)
-      plan.algo = results[0].algo;
-      if (found > 1) {
-        auto sx = at::empty({T, in}, x.options());
-        auto sy = at::empty({T, out}, x.options());
-        auto sw = at::zeros({out, in}, dw.options());
-        auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, x.options().dtype(at::kByte));
-        const float a1 = 1.0f, b1 = 1.0f;
-        auto stream = current_stream();
-        hipEvent_t ev0, ev1;
-        LPP_CHECK_HIP(hipEventCreate(&ev0));
-        LPP_CHECK_HIP(hipEventCreate(&ev1));
-        float best = 1e30f;
-        for (int i = 0; i < found; ++i) {
-          // one warm + one timed run per candidate
-          for (int rep = 0; rep < 2; ++rep) {
-            if (rep == 1) LPP_CHECK_HIP(hipEventRecord(ev0, stream));
-            hipblasStatus_t st = hipblasLtMatmul(
-                lt_ctx().handle, plan.op, &a1, sx.data_ptr(), plan.a, sy.data_ptr(),
-                plan.b, &b1, sw.data_ptr(), plan.c, sw.data_ptr(), plan.c,
-                &results[i].algo, wsbuf.data_ptr(), kWorkspaceBytes, stream);
-            if (st != HIPBLAS_STATUS_SUCCESS) { best = best; goto next_algo; }
-          }
-          LPP_CHECK_HIP(hipEventRecord(ev1, stream));
-          LPP_CHECK_HIP(hipEventSynchronize(ev1));
-          {
-            float ms = 0.f;
-            LPP_CHECK_HIP(hipEventElapsedTime(&ms, ev0, ev1));
-            if (ms < best) { best = ms; plan.algo = results[i].algo; }
-          }
-        next_algo:;
-        }
-        LPP_CHECK_HIP(hipEventDestroy(ev0));
-        LPP_CHECK_HIP(hipEventDestroy(ev1));
-      }
-      plans.emplace(key, plan);
-    }
-  }
-
+  CachedPlan plan = get_plan(T, in, out);
   auto workspace = at::empty({(int64_t)kWorkspaceBytes},
                              x.options().dtype(at::kByte));
   const float alpha = 1.0f, beta = 1.0f;
@@ -150,8 +185,86 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
       workspace.data_ptr(), kWorkspaceBytes, current_stream()));
 }
 
+// Exhaustive sweep: every library solution supported for this problem,
+// device-timed.  Returns (index, ms, kernel_name) sorted fastest-first.
+std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
+                                                                 int64_t out,
+                                                                 int64_t reps) {
+  CachedPlan plan = make_problem(T, in, out);
+  std::vector<hipblasLtMatmulHeuristicResult_t> all;
+  LPP_CHECK_BLASLT(hipblaslt_ext::getAllAlgos(
+      lt_ctx().handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM, HIPBLAS_OP_N,
+      HIPBLAS_OP_T, HIP_R_16BF, HIP_R_16BF, HIP_R_32F, HIP_R_32F, HIPBLAS_COMPUTE_32F,
+      all));
+  auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
+  auto sx = at::empty({T, in}, opt);
+  auto sy = at::empty({T, out}, opt);
+  auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
+  auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
+  auto stream = current_stream();
+  const float a1 = 1.0f, b1 = 1.0f;
+  std::vector<std::tuple<int64_t, double, std::string>> timed;
+  for (auto& cand : all) {
+    size_t ws_needed = 0;
+    hipblasStatus_t ok = hipblaslt_ext::matmulIsAlgoSupported(
+        lt_ctx().handle, plan.op, &a1, plan.a, plan.b, &b1, plan.c, plan.c, cand.algo,
+        ws_needed);
+    if (ok != HIPBLAS_STATUS_SUCCESS || ws_needed > kWorkspaceBytes) continue;
+    float ms = time_algo(plan, cand.algo, sx.data_ptr(), sy.data_ptr(), sw.data_ptr(),
+                         wsbuf.data_ptr(), (int)reps, stream);
+    if (ms >= 1e29f) continue;
+    int idx = hipblaslt_ext::getIndexFromAlgo(cand.algo);
+    timed.emplace_back(idx, (double)ms,
+                       hipblaslt_ext::getKernelNameFromAlgo(lt_ctx().handle, cand.algo));
+  }
+  std::sort(timed.begin(), timed.end(),
+            [](auto& a, auto& b) { return std::get<1>(a) < std::get<1>(b); });
+  return timed;
+}
+
+// Pin a committed solution index for a shape (overrides first-use heuristic).
+void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index) {
+  std::vector<int> want{(int)index};
+  std::vector<hipblasLtMatmulHeuristicResult_t> got;
+  LPP_CHECK_BLASLT(hipblaslt_ext::getAlgosFromIndex(lt_ctx().handle, want, got));
+  TORCH_CHECK(!got.empty(), "hipblaslt: no solution at index ", index);
+  CachedPlan plan = make_problem(T, in, out);
+  const float a1 = 1.0f, b1 = 1.0f;
+  size_t ws_needed = 0;
+  hipblasStatus_t ok = hipblaslt_ext::matmulIsAlgoSupported(
+      lt_ctx().handle, plan.op, &a1, plan.a, plan.b, &b1, plan.c, plan.c, got[0].algo,
+      ws_needed);
+  TORCH_CHECK(ok == HIPBLAS_STATUS_SUCCESS && ws_needed <= kWorkspaceBytes,
+              "hipblaslt: pinned algo ", index, " unsupported for wgrad [", out, ",",
+              in, "] k=", T);
+  plan.algo = got[0].algo;
+  std::lock_guard<std::mutex> lock(g_mu);
+  g_plans[AlgoKey{in, out, T}] = plan;
+}
+
+// What did the default path pick (for A/B reporting)?  Returns (index, name).
+std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out) {
+  CachedPlan plan = get_plan(T, in, out);
+  int idx = hipblaslt_ext::getIndexFromAlgo(plan.algo);
+  return {idx, hipblaslt_ext::getKernelNameFromAlgo(lt_ctx().handle, plan.algo)};
+}
+
 }  // namespace lpp
 
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   lpp::wgrad_f32_accum(x, dy, dw);
+}
+
+std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
+                                                                 int64_t out,
+                                                                 int64_t reps) {
+  return lpp::wgrad_tune(T, in, out, reps);
+}
+
+void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index) {
+  lpp::wgrad_set_algo(T, in, out, index);
+}
+
+std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out) {
+  return lpp::wgrad_current_algo(T, in, out);
 }
