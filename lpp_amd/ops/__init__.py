@@ -59,7 +59,8 @@ def _pin_wgrad_algos(ext) -> None:
     try:
         data = json.load(open(path))
         for entry in data.get("shapes", []):
-            ext.wgrad_set_algo(entry["T"], entry["in"], entry["out"], entry["index"])
+            ext.wgrad_set_algo(entry["T"], entry["in"], entry["out"], entry["index"],
+                               entry.get("kind", 0))
     except Exception as e:  # pragma: no cover
         import logging
 

@@ -20,11 +20,15 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
                                       at::Tensor v, at::Tensor o, at::Tensor lse2);
 at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw);
+void wgrad_f32_accum_pre(at::Tensor xT, at::Tensor dyT, at::Tensor dw);
+at::Tensor transpose2d(at::Tensor in);
 std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
                                                                  int64_t out,
-                                                                 int64_t reps);
-void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index);
-std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out);
+                                                                 int64_t reps,
+                                                                 int64_t kind);
+void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index, int64_t kind);
+std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out,
+                                                    int64_t kind);
 at::Tensor mfma_test_32x32x16(at::Tensor A, at::Tensor B);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> masters,
                  std::vector<at::Tensor> grads, std::vector<at::Tensor> exp_avg,
@@ -46,11 +50,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_fwd", &attention_fwd, "flash causal attention forward (O, LSE2)");
   m.def("attention_bwd", &attention_bwd, "flash causal attention backward (dQ, dK, dV)");
   m.def("wgrad_f32_accum", &wgrad_f32_accum, "dW_f32 += dY^T @ X (hipBLASLt, beta=1)");
+  m.def("wgrad_f32_accum_pre", &wgrad_f32_accum_pre,
+        "dW_f32 += dyT @ xT^T, pre-transposed k-contiguous operands (hipBLASLt, beta=1)");
+  m.def("transpose2d", &transpose2d, "bf16/fp16 [R,C] -> [C,R] LDS-tiled transpose");
   m.def("wgrad_tune", &wgrad_tune,
-        "exhaustive hipBLASLt solution sweep for a wgrad shape -> [(index, ms, name)]");
-  m.def("wgrad_set_algo", &wgrad_set_algo, "pin a hipBLASLt solution index for a shape");
+        "exhaustive hipBLASLt solution sweep for a wgrad shape -> [(index, ms, name)]",
+        py::arg("T"), py::arg("in_dim"), py::arg("out"), py::arg("reps"), py::arg("kind") = 0);
+  m.def("wgrad_set_algo", &wgrad_set_algo, "pin a hipBLASLt solution index for a shape",
+        py::arg("T"), py::arg("in_dim"), py::arg("out"), py::arg("index"), py::arg("kind") = 0);
   m.def("wgrad_current_algo", &wgrad_current_algo,
-        "current (heuristic-picked) solution for a shape -> (index, name)");
+        "current (heuristic-picked) solution for a shape -> (index, name)",
+        py::arg("T"), py::arg("in_dim"), py::arg("out"), py::arg("kind") = 0);
   m.def("mfma_test_16x16x32", &mfma_test_16x16x32, "MFMA layout validation");
   m.def("mfma_test_32x32x16", &mfma_test_32x32x16, "MFMA 32x32x16 layout validation");
 }

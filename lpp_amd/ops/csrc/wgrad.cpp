@@ -54,12 +54,15 @@ static LtContext& lt_ctx() {
 }
 
 struct AlgoKey {
-  int64_t m, n, k;
-  bool operator==(const AlgoKey& o) const { return m == o.m && n == o.n && k == o.k; }
+  int64_t m, n, k, kind;  // kind 0 = natural (k-strided), 1 = pre-transposed
+  bool operator==(const AlgoKey& o) const {
+    return m == o.m && n == o.n && k == o.k && kind == o.kind;
+  }
 };
 struct AlgoKeyHash {
   size_t operator()(const AlgoKey& k) const {
-    return std::hash<int64_t>()(k.m * 1315423911 ^ k.n * 2654435761 ^ k.k);
+    return std::hash<int64_t>()(k.m * 1315423911 ^ k.n * 2654435761 ^ k.k ^
+                                (k.kind << 60));
   }
 };
 
@@ -73,17 +76,35 @@ static std::mutex g_mu;
 static std::unordered_map<AlgoKey, CachedPlan, AlgoKeyHash> g_plans;
 
 // Create desc/layouts for the wgrad problem (no algo chosen yet).
-static CachedPlan make_problem(int64_t T, int64_t in, int64_t out) {
+//
+// kind 0 (natural): A = X rm[T,in], B = dY rm[T,out] — both operands
+//   contiguous along the OUTPUT dims, contraction T strided (the slow
+//   hipBLASLt class, ~1.0-1.2 PF/s at the 65B shapes).
+// kind 1 (pre-transposed): A = xT rm[in,T], B = dyT rm[out,T] — both
+//   contiguous along the contraction dim, the same class as the forward
+//   GEMM (~1.33-1.67 PF/s); operands produced by ops/csrc/transpose.hip.
+static CachedPlan make_problem(int64_t T, int64_t in, int64_t out, int kind) {
   CachedPlan plan;
   LPP_CHECK_BLASLT(hipblasLtMatmulDescCreate(&plan.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
   hipblasOperation_t opN = HIPBLAS_OP_N, opT = HIPBLAS_OP_T;
-  LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
-      plan.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opN, sizeof(opN)));
-  LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
-      plan.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opT, sizeof(opT)));
-  // cm views: A = X [in, T] ld=in; B = dY [out, T] ld=out (opB=T); C/D = [in, out] ld=in
-  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, in, T, in));
-  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, out, T, out));
+  if (kind == 0) {
+    LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+        plan.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opN, sizeof(opN)));
+    LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+        plan.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opT, sizeof(opT)));
+    // cm views: A = X [in, T] ld=in; B = dY [out, T] ld=out (opB=T); C/D = [in, out] ld=in
+    LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, in, T, in));
+    LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, out, T, out));
+  } else {
+    // cm views: A = xT [T, in] ld=T (opA=T); B = dyT [T, out] ld=T (opB=N);
+    // C/D = [in, out] ld=in  — both operands k(T)-contiguous
+    LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+        plan.op, HIPBLASLT_MATMUL_DESC_TRANSA, &opT, sizeof(opT)));
+    LPP_CHECK_BLASLT(hipblasLtMatmulDescSetAttribute(
+        plan.op, HIPBLASLT_MATMUL_DESC_TRANSB, &opN, sizeof(opN)));
+    LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, T, in, T));
+    LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, T, out, T));
+  }
   LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.c, HIP_R_32F, in, out, in));
   return plan;
 }
@@ -120,7 +141,8 @@ static float time_algo(CachedPlan& plan, hipblasLtMatmulAlgo_t& algo, void* ax, 
 }
 
 // Default first-use selection: heuristic top-48, each device-timed once.
-static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out) {
+static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out,
+                           int kind) {
   hipblasLtMatmulPreference_t pref;
   LPP_CHECK_BLASLT(hipblasLtMatmulPreferenceCreate(&pref));
   size_t ws = kWorkspaceBytes;
@@ -137,8 +159,8 @@ static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out)
   plan.algo = results[0].algo;
   if (found > 1) {
     auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
-    auto sx = at::empty({T, in}, opt);
-    auto sy = at::empty({T, out}, opt);
+    auto sx = kind == 0 ? at::empty({T, in}, opt) : at::empty({in, T}, opt);
+    auto sy = kind == 0 ? at::empty({T, out}, opt) : at::empty({out, T}, opt);
     auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
     auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
     auto stream = current_stream();
@@ -154,13 +176,13 @@ static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out)
   }
 }
 
-static CachedPlan get_plan(int64_t T, int64_t in, int64_t out) {
-  const AlgoKey key{in, out, T};
+static CachedPlan get_plan(int64_t T, int64_t in, int64_t out, int kind) {
+  const AlgoKey key{in, out, T, kind};
   std::lock_guard<std::mutex> lock(g_mu);
   auto it = g_plans.find(key);
   if (it != g_plans.end()) return it->second;
-  CachedPlan plan = make_problem(T, in, out);
-  pick_heuristic(plan, T, in, out);
+  CachedPlan plan = make_problem(T, in, out, kind);
+  pick_heuristic(plan, T, in, out, kind);
   g_plans.emplace(key, plan);
   return plan;
 }
@@ -175,7 +197,7 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   TORCH_CHECK(dy.size(0) == T && dw.size(0) == out && dw.size(1) == in,
               "wgrad shapes: x[T,in] dy[T,out] dw[out,in]");
 
-  CachedPlan plan = get_plan(T, in, out);
+  CachedPlan plan = get_plan(T, in, out, 0);
   auto workspace = at::empty({(int64_t)kWorkspaceBytes},
                              x.options().dtype(at::kByte));
   const float alpha = 1.0f, beta = 1.0f;
@@ -185,20 +207,42 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
       workspace.data_ptr(), kWorkspaceBytes, current_stream()));
 }
 
+// Pre-transposed variant: dW(rm [out,in], fp32) += dyT(rm [out,T])^rm-mm xT(rm [in,T])
+// — the k-contiguous hipBLASLt class (see make_problem kind 1).
+void wgrad_f32_accum_pre(at::Tensor xT, at::Tensor dyT, at::Tensor dw) {
+  TORCH_CHECK(xT.is_cuda() && xT.scalar_type() == at::kBFloat16 && xT.dim() == 2);
+  TORCH_CHECK(dyT.is_cuda() && dyT.scalar_type() == at::kBFloat16 && dyT.dim() == 2);
+  TORCH_CHECK(dw.is_cuda() && dw.scalar_type() == at::kFloat && dw.dim() == 2);
+  TORCH_CHECK(xT.is_contiguous() && dyT.is_contiguous() && dw.is_contiguous());
+  const int64_t in = xT.size(0), T = xT.size(1), out = dyT.size(0);
+  TORCH_CHECK(dyT.size(1) == T && dw.size(0) == out && dw.size(1) == in,
+              "wgrad_pre shapes: xT[in,T] dyT[out,T] dw[out,in]");
+
+  CachedPlan plan = get_plan(T, in, out, 1);
+  auto workspace = at::empty({(int64_t)kWorkspaceBytes},
+                             xT.options().dtype(at::kByte));
+  const float alpha = 1.0f, beta = 1.0f;
+  LPP_CHECK_BLASLT(hipblasLtMatmul(
+      lt_ctx().handle, plan.op, &alpha, xT.data_ptr(), plan.a, dyT.data_ptr(), plan.b,
+      &beta, dw.data_ptr(), plan.c, dw.data_ptr(), plan.c, &plan.algo,
+      workspace.data_ptr(), kWorkspaceBytes, current_stream()));
+}
+
 // Exhaustive sweep: every library solution supported for this problem,
 // device-timed.  Returns (index, ms, kernel_name) sorted fastest-first.
 std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
                                                                  int64_t out,
-                                                                 int64_t reps) {
-  CachedPlan plan = make_problem(T, in, out);
+                                                                 int64_t reps,
+                                                                 int64_t kind) {
+  CachedPlan plan = make_problem(T, in, out, (int)kind);
   std::vector<hipblasLtMatmulHeuristicResult_t> all;
   LPP_CHECK_BLASLT(hipblaslt_ext::getAllAlgos(
-      lt_ctx().handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM, HIPBLAS_OP_N,
-      HIPBLAS_OP_T, HIP_R_16BF, HIP_R_16BF, HIP_R_32F, HIP_R_32F, HIPBLAS_COMPUTE_32F,
-      all));
+      lt_ctx().handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
+      kind == 0 ? HIPBLAS_OP_N : HIPBLAS_OP_T, kind == 0 ? HIPBLAS_OP_T : HIPBLAS_OP_N,
+      HIP_R_16BF, HIP_R_16BF, HIP_R_32F, HIP_R_32F, HIPBLAS_COMPUTE_32F, all));
   auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
-  auto sx = at::empty({T, in}, opt);
-  auto sy = at::empty({T, out}, opt);
+  auto sx = kind == 0 ? at::empty({T, in}, opt) : at::empty({in, T}, opt);
+  auto sy = kind == 0 ? at::empty({T, out}, opt) : at::empty({out, T}, opt);
   auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
   auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
   auto stream = current_stream();
@@ -223,12 +267,12 @@ std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int6
 }
 
 // Pin a committed solution index for a shape (overrides first-use heuristic).
-void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index) {
+void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index, int64_t kind) {
   std::vector<int> want{(int)index};
   std::vector<hipblasLtMatmulHeuristicResult_t> got;
   LPP_CHECK_BLASLT(hipblaslt_ext::getAlgosFromIndex(lt_ctx().handle, want, got));
   TORCH_CHECK(!got.empty(), "hipblaslt: no solution at index ", index);
-  CachedPlan plan = make_problem(T, in, out);
+  CachedPlan plan = make_problem(T, in, out, (int)kind);
   const float a1 = 1.0f, b1 = 1.0f;
   size_t ws_needed = 0;
   hipblasStatus_t ok = hipblaslt_ext::matmulIsAlgoSupported(
@@ -239,12 +283,13 @@ void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index) {
               in, "] k=", T);
   plan.algo = got[0].algo;
   std::lock_guard<std::mutex> lock(g_mu);
-  g_plans[AlgoKey{in, out, T}] = plan;
+  g_plans[AlgoKey{in, out, T, kind}] = plan;
 }
 
 // What did the default path pick (for A/B reporting)?  Returns (index, name).
-std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out) {
-  CachedPlan plan = get_plan(T, in, out);
+std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out,
+                                                    int64_t kind) {
+  CachedPlan plan = get_plan(T, in, out, (int)kind);
   int idx = hipblaslt_ext::getIndexFromAlgo(plan.algo);
   return {idx, hipblaslt_ext::getKernelNameFromAlgo(lt_ctx().handle, plan.algo)};
 }
@@ -255,16 +300,22 @@ void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   lpp::wgrad_f32_accum(x, dy, dw);
 }
 
+void wgrad_f32_accum_pre(at::Tensor xT, at::Tensor dyT, at::Tensor dw) {
+  lpp::wgrad_f32_accum_pre(xT, dyT, dw);
+}
+
 std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
                                                                  int64_t out,
-                                                                 int64_t reps) {
-  return lpp::wgrad_tune(T, in, out, reps);
+                                                                 int64_t reps,
+                                                                 int64_t kind) {
+  return lpp::wgrad_tune(T, in, out, reps, kind);
 }
 
-void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index) {
-  lpp::wgrad_set_algo(T, in, out, index);
+void wgrad_set_algo(int64_t T, int64_t in, int64_t out, int64_t index, int64_t kind) {
+  lpp::wgrad_set_algo(T, in, out, index, kind);
 }
 
-std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out) {
-  return lpp::wgrad_current_algo(T, in, out);
+std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64_t out,
+                                                    int64_t kind) {
+  return lpp::wgrad_current_algo(T, in, out, kind);
 }

@@ -60,9 +60,21 @@ def main():
             dgrad = timeit(lambda: torch.matmul(dy, w))
             wg_bf16 = timeit(lambda: torch.matmul(dy.t(), x))
             wg_f32 = timeit(lambda: ext.wgrad_f32_accum(x, dy, dw32))
+            # pre-transposed operands turn the wgrad into the k-contiguous
+            # class (same as fwd): dW[out,in] = dyT[out,T] @ xT[in,T]^T
+            xT = x.t().contiguous()
+            dyT = dy.t().contiguous()
+            wg_pre = timeit(lambda: torch.matmul(dyT, xT.t()))
+            tx = timeit(lambda: x.t().contiguous())
+            tdy = timeit(lambda: dy.t().contiguous())
+            xbytes = 2 * 2.0 * T * cin / 1e9
+            dybytes = 2 * 2.0 * T * cout / 1e9
             print(f"[{name} T={T} in={cin} out={cout}] "
                   f"fwd(NT) {tf(fwd):5.0f}  dgrad(NN) {tf(dgrad):5.0f}  "
-                  f"wgradTN(bf16D) {tf(wg_bf16):5.0f}  wgradTN(f32D,b1) {tf(wg_f32):5.0f} TF/s",
+                  f"wgradTN(bf16D) {tf(wg_bf16):5.0f}  wgradTN(f32D,b1) {tf(wg_f32):5.0f}  "
+                  f"wgradPRE(bf16D) {tf(wg_pre):5.0f} TF/s  "
+                  f"| torch-transpose x {tx:6.3f} ms ({xbytes / (tx * 1e-3):4.0f} GB/s) "
+                  f"dy {tdy:6.3f} ms ({dybytes / (tdy * 1e-3):4.0f} GB/s)",
                   flush=True)
 
 

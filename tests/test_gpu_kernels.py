@@ -239,3 +239,72 @@ def test_rmsnorm_bwd_deterministic_dw(ext):
     _, dw1 = ext.rmsnorm_bwd(dy, x, w, inv)
     _, dw2 = ext.rmsnorm_bwd(dy, x, w, inv)
     assert torch.equal(dw1, dw2)
+
+
+def test_transpose2d(ext):
+    """LDS-tiled transpose vs torch .t() for even and ragged shapes."""
+    torch.manual_seed(21)
+    for R, C in ((512, 256), (4096, 8192), (100, 130), (64, 72)):
+        x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16)
+        assert torch.equal(ext.transpose2d(x), x.t().contiguous()), (R, C)
+    h = torch.randn(256, 192, device="cuda", dtype=torch.float16)
+    assert torch.equal(ext.transpose2d(h), h.t().contiguous())
+
+
+def test_wgrad_f32_accum_pre(ext):
+    """Pre-transposed wgrad formulation == natural formulation == fp32 ref."""
+    torch.manual_seed(22)
+    T, IN, OUT = 512, 256, 384
+    x = torch.randn(T, IN, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(T, OUT, device="cuda", dtype=torch.bfloat16)
+    dw = torch.randn(OUT, IN, device="cuda", dtype=torch.float32)
+    ref = dw + dy.float().t() @ x.float()
+    ext.wgrad_f32_accum_pre(ext.transpose2d(x), ext.transpose2d(dy), dw)
+    err = (dw - ref).abs().max() / ref.abs().max().clamp(min=1)
+    assert err < 2e-2, err
+
+
+def test_lp_linear_pre_path_matches_natural(monkeypatch):
+    """LPLinear backward with LPP_WGRAD_PRE on/off accumulates the same
+    main_grad (down to GEMM reduction-order noise)."""
+    from lpp_amd.ops.linear import lp_linear
+
+    torch.manual_seed(23)
+    B, S, IN, OUT = 2, 128, 256, 192
+
+    def run(pre: str):
+        monkeypatch.setenv("LPP_WGRAD_PRE", pre)
+        torch.manual_seed(23)
+        w = torch.nn.Parameter(
+            torch.randn(OUT, IN, device="cuda", dtype=torch.bfloat16) * 0.05
+        )
+        x = torch.randn(B, S, IN, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        dy = torch.randn(B, S, OUT, device="cuda", dtype=torch.bfloat16)
+        w.main_grad = torch.zeros(OUT, IN, device="cuda", dtype=torch.float32)
+        out = lp_linear(x, w)
+        out.backward(dy)
+        return w.main_grad.clone(), x.grad.clone()
+
+    g_pre, dx_pre = run("1")
+    g_nat, dx_nat = run("0")
+    assert torch.equal(dx_pre, dx_nat)
+    err = (g_pre - g_nat).abs().max() / g_nat.abs().max().clamp(min=1e-3)
+    assert err < 1e-2, err
+
+
+def test_xt_cache_shared_and_invalidated(ext):
+    """The transpose memo reuses xT for the same activation and refreshes
+    for a new one."""
+    from lpp_amd.ops import linear as L
+
+    L._xt_cache.clear()
+    x1 = torch.randn(64, 128, device="cuda", dtype=torch.bfloat16)
+    t1 = L._xt_cache.get(ext, x1)
+    t1b = L._xt_cache.get(ext, x1)
+    assert t1 is t1b
+    x2 = torch.randn(64, 128, device="cuda", dtype=torch.bfloat16)
+    t2 = L._xt_cache.get(ext, x2)
+    assert t2 is not t1
+    assert torch.equal(t2, x2.t().contiguous())
+    L._xt_cache.clear()
