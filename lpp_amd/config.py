@@ -170,6 +170,17 @@ class TrainConfig:
     # Data
     total_dataset_len: int = 0  # quirk Q3 fix: broadcast once (see engine)
     data_pattern: str = "uniform"  # synthetic data: uniform | arith (learnable)
+    # Real-corpus training (reference: Hydra-instantiated dataset/collator,
+    # trainer_base_ds_mp.py:142-200,317).  data_kind "synthetic" needs no
+    # files; "jsonl" trains on train_file ({"inputs","targets"} records —
+    # .jsonl/.json/torch-saved list) through Seq2SeqToCausalLM/TextCollator.
+    data_kind: str = "synthetic"  # synthetic | jsonl
+    train_file: str = ""
+    # tokenizer: "simple" = offline whitespace SimpleTokenizer; otherwise a
+    # HF tokenizer directory (defaults to model_name_or_path when that is a
+    # converted checkpoint dir carrying tokenizer files)
+    tokenizer_path: str = ""
+    data_field: Optional[str] = None  # unwrap nested mixing-dataset items
     # shell command run by rank 0 after each checkpoint save, with {dir}
     # substituted (reference: ./s5cmd sync to S3, trainer_base_ds_mp.py:220)
     save_hook_cmd: str = ""

@@ -38,12 +38,48 @@ from .utils import init_distributed, set_seed
 logger = logging.getLogger("lpp_amd.trainer")
 
 
+def build_tokenizer(cfg: TrainConfig):
+    """Tokenizer for the jsonl path: a HF tokenizer directory when given (or
+    found inside model_name_or_path, as convert2ckpt passes tokenizer files
+    through), else the offline SimpleTokenizer.  Either way
+    expand_special_tokenizer applies (reference trainer_base_ds_mp.py:416-420)."""
+    from .data import SimpleTokenizer, expand_special_tokenizer
+
+    path = cfg.tokenizer_path
+    if not path and cfg.model_name_or_path and any(
+        os.path.exists(os.path.join(cfg.model_name_or_path, f))
+        for f in ("tokenizer.json", "tokenizer.model", "tokenizer_config.json")
+    ):
+        path = cfg.model_name_or_path
+    if path and path != "simple":
+        from transformers import AutoTokenizer
+
+        tok = AutoTokenizer.from_pretrained(path)
+    else:
+        tok = SimpleTokenizer(cfg.model.vocab_size)
+    expand_special_tokenizer(tok)
+    return tok
+
+
 def build_dataset(cfg: TrainConfig):
-    """Synthetic dataset by default (no-network environment); a real corpus
-    drops in behind the same dict contract."""
-    n = cfg.total_dataset_len or 4096
-    return SyntheticCausalLMDataset(n, cfg.seq_len, cfg.model.vocab_size, seed=cfg.seed,
-                                    pattern=getattr(cfg, "data_pattern", "uniform"))
+    """Config-selected (dataset, collator) — the native version of the
+    reference's Hydra dataset/collator instantiation
+    (trainer_base_ds_mp.py:142-200,317).  Returns (dataset, collator)."""
+    if cfg.data_kind == "synthetic":
+        n = cfg.total_dataset_len or 4096
+        ds = SyntheticCausalLMDataset(n, cfg.seq_len, cfg.model.vocab_size,
+                                      seed=cfg.seed,
+                                      pattern=getattr(cfg, "data_pattern", "uniform"))
+        return ds, CausalLMCollator(cfg.seq_len)
+    if cfg.data_kind == "jsonl":
+        from .data import PromptResponseDataset, TextCollator
+
+        if not cfg.train_file:
+            raise ValueError("data_kind=jsonl requires train_file")
+        tok = build_tokenizer(cfg)
+        ds = PromptResponseDataset(cfg.train_file)
+        return ds, TextCollator(tok, cfg.seq_len, field=cfg.data_field)
+    raise ValueError(f"unknown data_kind {cfg.data_kind!r} (synthetic | jsonl)")
 
 
 def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dict:
@@ -62,11 +98,25 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
             wandb = None
 
     needs_data = grid.is_first_stage() or grid.is_last_stage()
-    dataset = build_dataset(cfg) if needs_data else None
+    dataset = collator = None
+    # rank-0-first build (reference barrier discipline,
+    # trainer_base_ds_mp.py:163-176): rank 0 pays any tokenizer/cache cost
+    # once; the other data-holding ranks hit the warm cache.
+    from .utils import rank_zero_first
 
-    # quirk Q3 fix: every rank derives the identical step count from config,
-    # not from its local dataloader length.
-    n_examples = cfg.total_dataset_len or (len(dataset) if dataset is not None else 4096)
+    with rank_zero_first(grid.rank):
+        if needs_data:
+            dataset, collator = build_dataset(cfg)
+
+    # quirk Q3 fix: every rank uses the IDENTICAL example count — from
+    # config when pinned, else measured on rank 0 and broadcast (middle
+    # stages hold no dataset and must not guess).
+    n_examples = cfg.total_dataset_len or (len(dataset) if dataset is not None else 0)
+    if dist.is_initialized():
+        obj = [n_examples]
+        dist.broadcast_object_list(obj, src=0)
+        n_examples = int(obj[0])
+    n_examples = n_examples or 4096
     examples_per_step = cfg.micro_batch_size * cfg.gradient_accumulation_steps * grid.dp_degree
     steps_per_epoch = n_examples // examples_per_step
     total_steps = cfg.max_steps or steps_per_epoch * cfg.num_train_epochs
@@ -94,7 +144,7 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
             loader = build_loader(
                 dataset, cfg.micro_batch_size, grid.dp_degree, grid.dp_id,
                 seed=cfg.seed, num_workers=cfg.num_workers,
-                collator=CausalLMCollator(cfg.seq_len), epoch=epoch,
+                collator=collator, epoch=epoch,
             )
             it = iter(RepeatingLoader(loader))
             if cfg.eval_steps:
@@ -104,7 +154,7 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
                 eval_loader = build_loader(
                     dataset, cfg.micro_batch_size, grid.dp_degree, grid.dp_id,
                     seed=cfg.seed + 7919, num_workers=0,
-                    collator=CausalLMCollator(cfg.seq_len), epoch=epoch,
+                    collator=collator, epoch=epoch,
                 )
                 eval_it = iter(RepeatingLoader(eval_loader))
         for _ in range(steps_per_epoch):

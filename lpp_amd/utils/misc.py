@@ -57,6 +57,34 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 7200) -> tu
     return dist.get_rank(), dist.get_world_size()
 
 
+import contextlib
+
+
+@contextlib.contextmanager
+def rank_zero_first(rank: int = None):
+    """Rank-0-first execution discipline for dataset building/caching: rank 0
+    runs the body while the others wait at a barrier, then the others run it
+    (hitting the cache rank 0 produced) and everyone re-syncs — the
+    reference's barrier pattern at trainer_base_ds_mp.py:163-176.  Barriers
+    must be matched on ALL ranks, so call this on every rank."""
+    if not dist.is_initialized():
+        yield
+        return
+    if rank is None:
+        rank = dist.get_rank()
+    # barrier A: non-zero ranks wait in it while rank 0 runs the body;
+    # rank 0 enters it after the body, releasing them.
+    if rank != 0:
+        dist.barrier()
+    try:
+        yield
+    finally:
+        if rank == 0:
+            dist.barrier()  # barrier A (release)
+        # barrier B: everyone re-syncs after their own pass over the body
+        dist.barrier()
+
+
 class StepTimer:
     """Cheap wall-clock step timer with device sync on CUDA."""
 
