@@ -145,6 +145,9 @@ def load_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
     # refresh masters from (re)loaded params (mode-aware: ZeRO-1 keeps one
     # flat master shard, not per-param copies)
     engine.optimizer.refresh_masters()
+    from .ops.linear import invalidate_weight_transposes
+
+    invalidate_weight_transposes(engine.module)
 
     client_state = None
     if not load_module_only:

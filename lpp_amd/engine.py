@@ -570,6 +570,11 @@ class PipelineEngine:
             self.optimizer.zero_grad()
             self.lr_scheduler.step()
             self.global_steps += 1
+            # weights changed: drop the cached W^T dgrad operands
+            # (lazily rebuilt by the first backward of the next step)
+            from .ops.linear import invalidate_weight_transposes
+
+            invalidate_weight_transposes(self.module)
 
     # ------------------------------------------------------------------
     @property

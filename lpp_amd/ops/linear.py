@@ -55,6 +55,34 @@ def _wgrad_pre_enabled() -> bool:
     return os.environ.get("LPP_WGRAD_PRE", "1") == "1"
 
 
+def _dgrad_wt_enabled() -> bool:
+    return os.environ.get("LPP_DGRAD_WT", "1") == "1"
+
+
+def _weight_t(ext, weight: torch.nn.Parameter) -> torch.Tensor:
+    """Cached transposed weight for the dgrad GEMM.
+
+    dx = dy @ W is the k-strided hipBLASLt class (the contraction dim runs
+    down W's rows) and measures 8-21% below the k-contiguous class at the
+    65B shapes (profiles/r02_gemm_probe2.txt).  W only changes at the
+    optimizer step, so one transposed copy per step turns every dgrad into
+    the fast class:  dx = dy @ (W^T)^T with W^T materialised [in, out].
+    The engine invalidates the cache after each optimizer step
+    (engine._optimizer_step -> invalidate_weight_transposes)."""
+    wt = getattr(weight, "_wt", None)
+    if wt is None:
+        wt = ext.transpose2d(weight.data)
+        weight._wt = wt
+    return wt
+
+
+def invalidate_weight_transposes(module: torch.nn.Module) -> None:
+    """Drop cached W^T copies (call after any in-place weight update)."""
+    for p in module.parameters():
+        if hasattr(p, "_wt"):
+            p._wt = None
+
+
 class _LinearWgradF32(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.nn.Parameter):
@@ -68,8 +96,11 @@ class _LinearWgradF32(torch.autograd.Function):
     def backward(ctx, dy: torch.Tensor):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = dy @ weight
         ext = extension()
+        if _dgrad_wt_enabled():
+            dx = dy @ _weight_t(ext, weight).t()
+        else:
+            dx = dy @ weight
         x2 = x.reshape(-1, x.shape[-1]).contiguous()
         dy2 = dy.reshape(-1, dy.shape[-1])
         if _wgrad_pre_enabled():

@@ -19,8 +19,12 @@ from __future__ import annotations
 
 from dataclasses import dataclass
 
-HBM_BUDGET_BYTES = 248e9  # leave ~40 GB for logits/p2p/allocator slack
-BYTES_PER_PARAM = 18      # bf16 param + fp32 master/exp_avg/exp_avg_sq/grad
+# Leave ~24 GB of the 288 GB HBM3E for logits/p2p/allocator slack.  The
+# byte estimator itself is ~10 GB conservative vs measured peaks (r01/r02
+# bench peak_mem_gb), so the real slack is ~34 GB at the 65B shapes.
+HBM_BUDGET_BYTES = 264e9
+BYTES_PER_PARAM = 20      # bf16 param + fp32 master/exp_avg/exp_avg_sq/grad
+                          # + bf16 W^T dgrad copy (ops/linear._weight_t)
 
 
 @dataclass

@@ -308,3 +308,23 @@ def test_xt_cache_shared_and_invalidated(ext):
     assert t2 is not t1
     assert torch.equal(t2, x2.t().contiguous())
     L._xt_cache.clear()
+
+
+def test_dgrad_wt_cache_and_invalidate(ext):
+    """dx via the cached W^T equals dy @ W; invalidation refreshes after an
+    in-place weight update."""
+    from lpp_amd.ops import linear as L
+
+    w = torch.nn.Parameter(torch.randn(96, 128, device="cuda", dtype=torch.bfloat16))
+    dy = torch.randn(32, 96, device="cuda", dtype=torch.bfloat16)
+    wt = L._weight_t(ext, w)
+    assert torch.equal(wt, w.data.t().contiguous())
+    assert L._weight_t(ext, w) is wt  # cached
+    dx_ref = dy @ w
+    dx = dy @ wt.t()
+    assert torch.allclose(dx.float(), dx_ref.float(), rtol=2e-2, atol=1e-2)
+    with torch.no_grad():
+        w.data.mul_(2.0)
+    L.invalidate_weight_transposes(torch.nn.ParameterList([w]))
+    wt2 = L._weight_t(ext, w)
+    assert torch.equal(wt2, w.data.t().contiguous())
