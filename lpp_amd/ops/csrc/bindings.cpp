@@ -22,6 +22,7 @@ at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw);
 void wgrad_f32_accum_pre(at::Tensor xT, at::Tensor dyT, at::Tensor dw);
 at::Tensor transpose2d(at::Tensor in);
+void accum_bf16_f32(at::Tensor dst, at::Tensor src);
 std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
                                                                  int64_t out,
                                                                  int64_t reps,
@@ -53,6 +54,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_f32_accum_pre", &wgrad_f32_accum_pre,
         "dW_f32 += dyT @ xT^T, pre-transposed k-contiguous operands (hipBLASLt, beta=1)");
   m.def("transpose2d", &transpose2d, "bf16/fp16 [R,C] -> [C,R] LDS-tiled transpose");
+  m.def("accum_bf16_f32", &accum_bf16_f32, "dst_f32 += src_bf16 (vectorised)");
   m.def("wgrad_tune", &wgrad_tune,
         "exhaustive hipBLASLt solution sweep for a wgrad shape -> [(index, ms, name)]",
         py::arg("T"), py::arg("in_dim"), py::arg("out"), py::arg("reps"), py::arg("kind") = 0);

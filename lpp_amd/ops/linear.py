@@ -55,6 +55,16 @@ def _wgrad_pre_enabled() -> bool:
     return os.environ.get("LPP_WGRAD_PRE", "1") == "1"
 
 
+def _wgrad_bf16d_enabled() -> bool:
+    """bf16-D wgrad GEMM + separate fp32 accumulate (default): hipBLASLt's
+    bf16-D solution pool is 15-20% faster than fp32-D at the 65B shapes and
+    the accumulate pass is ~5% of the GEMM.  Numerics match the reference
+    stack's own flow (bf16 dW per microbatch, fp32 accumulation across
+    microbatches — SURVEY.md §2.5).  Set LPP_WGRAD_BF16D=0 for the exact
+    fp32-D GEMM epilogue."""
+    return os.environ.get("LPP_WGRAD_BF16D", "1") == "1"
+
+
 def _dgrad_wt_enabled() -> bool:
     return os.environ.get("LPP_DGRAD_WT", "1") == "1"
 
@@ -109,7 +119,12 @@ class _LinearWgradF32(torch.autograd.Function):
             # layout) at the cost of two HBM-speed LDS-tiled transposes;
             # the x transpose is shared across sibling projections.
             xT = _xt_cache.get(ext, x2)
-            ext.wgrad_f32_accum_pre(xT, ext.transpose2d(dy2), ctx.main_grad)
+            dyT = ext.transpose2d(dy2)
+            if _wgrad_bf16d_enabled():
+                ext.accum_bf16_f32(ctx.main_grad.view(-1),
+                                   torch.matmul(dyT, xT.t()).view(-1))
+            else:
+                ext.wgrad_f32_accum_pre(xT, dyT, ctx.main_grad)
         else:
             ext.wgrad_f32_accum(x2, dy2, ctx.main_grad)
         return dx, None

@@ -328,3 +328,36 @@ def test_dgrad_wt_cache_and_invalidate(ext):
     L.invalidate_weight_transposes(torch.nn.ParameterList([w]))
     wt2 = L._weight_t(ext, w)
     assert torch.equal(wt2, w.data.t().contiguous())
+
+
+def test_accum_bf16_f32(ext):
+    torch.manual_seed(31)
+    for n in (8 * 1024, 1000, 7):
+        dst = torch.randn(n, device="cuda", dtype=torch.float32)
+        src = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        ref = dst + src.float()
+        ext.accum_bf16_f32(dst, src)
+        assert torch.equal(dst, ref), n
+
+
+def test_wgrad_bf16d_matches_f32d(monkeypatch):
+    """bf16-D wgrad + fp32 accumulate stays within one bf16 rounding of the
+    exact fp32-D epilogue."""
+    from lpp_amd.ops.linear import lp_linear
+
+    def run(flag):
+        monkeypatch.setenv("LPP_WGRAD_BF16D", flag)
+        torch.manual_seed(33)
+        w = torch.nn.Parameter(
+            torch.randn(96, 128, device="cuda", dtype=torch.bfloat16) * 0.05)
+        x = torch.randn(4, 64, 128, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        dy = torch.randn(4, 64, 96, device="cuda", dtype=torch.bfloat16)
+        w.main_grad = torch.zeros(96, 128, device="cuda", dtype=torch.float32)
+        lp_linear(x, w).backward(dy)
+        return w.main_grad.clone()
+
+    g16 = run("1")
+    g32 = run("0")
+    err = (g16 - g32).abs().max() / g32.abs().max().clamp(min=1e-3)
+    assert err < 1e-2, err
