@@ -53,15 +53,28 @@ def main() -> int:
               f"{c.get('ckpt_layers_per_stage', '?'):>4}  {c.get('peak_mem_gb', '?')}")
         timers = c.get("stage_timers") or []
         if len(timers) > 1:
+            # stage_timers are DEVICE-measured (hipEvent pairs on the compute
+            # stream, engine.device_timers): fwd/bwd = kernel time, p2p /
+            # allreduce = stalls the compute stream suffered waiting on the
+            # comm channels.  p2p on rank 0/last ~= fill-drain bubble; on
+            # middle ranks ~= upstream stall.
             steps = r.get("steps", 1)
             worst = max(timers, key=lambda t: t["p2p_s"])
             total = r["ms_per_step"] * steps / 1000.0
-            print(f"     per-rank p2p wait: max {worst['p2p_s']:.1f}s "
+            print(f"     per-rank p2p stall (device-true): max {worst['p2p_s']:.1f}s "
                   f"(rank {worst['rank']}, {100 * worst['p2p_s'] / max(total, 1e-9):.0f}% "
-                  f"of the timed window) — p2p_s on rank 0/last ~= pipeline "
-                  f"fill/drain bubble; on middle ranks ~= upstream stall")
+                  f"of the timed window)")
+            busiest = max(timers, key=lambda t: t["fwd_s"] + t["bwd_s"])
+            idle = total - (busiest["fwd_s"] + busiest["bwd_s"] +
+                            busiest["p2p_s"] + busiest["allreduce_s"] +
+                            busiest["optim_s"])
+            print(f"     busiest rank {busiest['rank']}: compute "
+                  f"{busiest['fwd_s'] + busiest['bwd_s']:.1f}s, unattributed "
+                  f"{idle:.1f}s ({100 * idle / max(total, 1e-9):.0f}% — host gaps "
+                  f"/ dataloader if large)")
+            ar = max(t["allreduce_s"] for t in timers)
             opt = max(t["optim_s"] for t in timers)
-            print(f"     optimizer: max {opt:.1f}s across ranks")
+            print(f"     allreduce stall max {ar:.1f}s; optimizer max {opt:.1f}s")
     mbs_note = {r.get('n_gpus'): r.get('config', {}).get('micro_batch_size') for r in rows}
     if len(set(mbs_note.values())) > 1:
         print("note: microbatch size varies by N (adaptive policy) — efficiency "

@@ -147,6 +147,41 @@ def bench_attention(B=1, S=4096, H=64, D=128):
     return rows
 
 
+def bench_transpose(R=16384, C=8192):
+    from lpp_amd import ops
+
+    ext = ops.extension()
+    x = torch.randn(R, C, device=DEV, dtype=torch.bfloat16)
+    bytes_moved = x.numel() * 2 * 2
+    t_hip = timeit(lambda: ext.transpose2d(x))
+    t_ref = timeit(lambda: x.t().contiguous())
+    return [(f"transpose2d[{R}x{C}]", t_hip, t_ref, bytes_moved)]
+
+
+def bench_wgrad(T=16384, IN=8192, OUT=8192):
+    """wgrad formulations: eager = natural TN fp32-accum two-pass; hip =
+    the production pre-transposed bf16-D GEMM + fused fp32 accumulate
+    (includes its transpose costs)."""
+    from lpp_amd import ops
+
+    ext = ops.extension()
+    x = torch.randn(T, IN, device=DEV, dtype=torch.bfloat16)
+    dy = torch.randn(T, OUT, device=DEV, dtype=torch.bfloat16)
+    dw = torch.zeros(OUT, IN, device=DEV, dtype=torch.float32)
+
+    def hip():
+        xT = ext.transpose2d(x)
+        dyT = ext.transpose2d(dy)
+        ext.accum_bf16_f32(dw.view(-1), torch.matmul(dyT, xT.t()).view(-1))
+
+    def ref():
+        dw.add_(torch.matmul(dy.t().float(), x.float()))
+
+    t_hip = timeit(hip, iters=10)
+    t_ref = timeit(ref, iters=3)
+    return [(f"wgrad[{T}x{IN}->{OUT}]", t_hip, t_ref, 0)]
+
+
 def bench_adamw(n=1_000_000_000 // 4):
     from lpp_amd import ops
 
@@ -186,7 +221,7 @@ def main():
     torch.cuda.set_device(0)
     rows = []
     for fn in [bench_rmsnorm, bench_rmsnorm_bwd, bench_rope, bench_swiglu, bench_ce,
-               bench_adamw, bench_attention]:
+               bench_adamw, bench_attention, bench_transpose, bench_wgrad]:
         try:
             rows += fn()
         except Exception as e:
