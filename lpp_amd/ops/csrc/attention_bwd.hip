@@ -155,6 +155,10 @@ __global__ __launch_bounds__(256) void attn_bwd_delta_kernel(
 // double-buffered K (normal), V (normal), K^T (tswz image).
 constexpr int DQ_QW = 32, DQ_WAVES = 8, DQ_QB = 256, DQ_KVB = 64;
 
+// DBUF=true: double-buffered images (96 KB LDS, 1 block/CU, one barrier
+// per tile).  DBUF=false: single-buffered (48 KB, 2 blocks/CU, two
+// barriers per tile) — staging stalls hidden by the co-resident block.
+template <bool DBUF>
 __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
@@ -162,10 +166,13 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
     short* __restrict__ dQ,
     int B, int S, int H, int HKV, float c, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // K0@0 K1@16K V0@32K V1@48K Kt0@64K Kt1@64K+18432
-  auto k_lds = [&](int buf) -> char* { return smem + buf * 16384; };
-  auto v_lds = [&](int buf) -> char* { return smem + 32768 + buf * 16384; };
-  auto kt_lds = [&](int buf) -> char* { return smem + 65536 + buf * 16384; };
+  constexpr int NB = DBUF ? 2 : 1;
+  // K@0 V@NB*16K Kt@2*NB*16K (+buf*16K each when double-buffered)
+  auto k_lds = [&](int buf) -> char* { return smem + (DBUF ? buf : 0) * 16384; };
+  auto v_lds = [&](int buf) -> char* {
+    return smem + NB * 16384 + (DBUF ? buf : 0) * 16384; };
+  auto kt_lds = [&](int buf) -> char* {
+    return smem + 2 * NB * 16384 + (DBUF ? buf : 0) * 16384; };
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -291,8 +298,9 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
     }
 
     if (have_next) {
-      write_tile(cur ^ 1, 0, kreg0, vreg0);
-      write_tile(cur ^ 1, 1, kreg1, vreg1);
+      if (!DBUF) __syncthreads();  // everyone done reading the only buffer
+      write_tile(DBUF ? (cur ^ 1) : 0, 0, kreg0, vreg0);
+      write_tile(DBUF ? (cur ^ 1) : 0, 1, kreg1, vreg1);
     }
     __syncthreads();
   }
@@ -350,6 +358,7 @@ __device__ __forceinline__ bf16x8 pack_frag16(const float p[2][4], int hi4_odd) 
   return *reinterpret_cast<const bf16x8*>(w);
 }
 
+template <bool DBUF>
 __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
@@ -357,15 +366,21 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     short* __restrict__ dK, short* __restrict__ dV,
     int B, int S, int H, int HKV, float c, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // 64-row double-buffered tiles:
-  // Q0@0 Q1@16K dO0@32K dO1@48K Qt0@64K Qt1@+18432 dOt0@100864... all 16-B
-  // aligned: Qt/dOt are [128][72]*2B = 18432 B each.
-  auto q_lds = [&](int buf) -> char* { return smem + buf * 16384; };
-  auto do_lds = [&](int buf) -> char* { return smem + 32768 + buf * 16384; };
-  auto qt_lds = [&](int buf) -> char* { return smem + 65536 + buf * 16384; };
-  auto dot_lds = [&](int buf) -> char* { return smem + 98304 + buf * 16384; };
-  auto l_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 131072) + buf * 64; };
-  auto d_buf = [&](int buf) -> float* { return reinterpret_cast<float*>(smem + 131584) + buf * 64; };
+  constexpr int NB = DBUF ? 2 : 1;
+  // 64-row tiles, [128][64] transposed images (16 KB each):
+  // Q@0 dO@NB*16K Qt@2*NB*16K dOt@3*NB*16K L/D@4*NB*16K
+  auto q_lds = [&](int buf) -> char* { return smem + (DBUF ? buf : 0) * 16384; };
+  auto do_lds = [&](int buf) -> char* {
+    return smem + NB * 16384 + (DBUF ? buf : 0) * 16384; };
+  auto qt_lds = [&](int buf) -> char* {
+    return smem + 2 * NB * 16384 + (DBUF ? buf : 0) * 16384; };
+  auto dot_lds = [&](int buf) -> char* {
+    return smem + 3 * NB * 16384 + (DBUF ? buf : 0) * 16384; };
+  auto l_buf = [&](int buf) -> float* {
+    return reinterpret_cast<float*>(smem + 4 * NB * 16384) + (DBUF ? buf : 0) * 64; };
+  auto d_buf = [&](int buf) -> float* {
+    return reinterpret_cast<float*>(smem + 4 * NB * 16384 + NB * 256) +
+           (DBUF ? buf : 0) * 64; };
 
 
   const int tid = threadIdx.x;
@@ -511,8 +526,9 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
       }
 
       if (have_next) {
-        write_tile(cur ^ 1, qt0 + KV_QT, 0, qreg0, dreg0);
-        write_tile(cur ^ 1, qt0 + KV_QT, 1, qreg1, dreg1);
+        if (!DBUF) __syncthreads();  // everyone done reading the only buffer
+        write_tile(DBUF ? (cur ^ 1) : 0, qt0 + KV_QT, 0, qreg0, dreg0);
+        write_tile(DBUF ? (cur ^ 1) : 0, qt0 + KV_QT, 1, qreg1, dreg1);
       }
       __syncthreads();
     }
@@ -562,10 +578,18 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
 
+  // LPP_ATTN_DBUF=1 restores the double-buffered (1 block/CU) variants;
+  // default is single-buffered at 2 blocks/CU (co-resident block hides the
+  // staging barriers).
+  static const bool dbuf = [] {
+    const char* e = getenv("LPP_ATTN_DBUF");
+    return e && e[0] == '1';
+  }();
   {
     const int qblocks = (S + lpp::DQ_QB - 1) / lpp::DQ_QB;
-    const size_t lds = 98304;
-    hipLaunchKernelGGL(lpp::attn_bwd_dq_kernel, dim3(qblocks, B * H), dim3(512), lds,
+    const size_t lds = dbuf ? 98304 : 49152;
+    auto kfn = dbuf ? lpp::attn_bwd_dq_kernel<true> : lpp::attn_bwd_dq_kernel<false>;
+    hipLaunchKernelGGL(kfn, dim3(qblocks, B * H), dim3(512), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
                        lse2.data_ptr<float>(), delta.data_ptr<float>(),
@@ -574,8 +598,9 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
   }
   {
     const int kvblocks = (S + lpp::KV_WG - 1) / lpp::KV_WG;
-    const size_t lds = 132096;
-    hipLaunchKernelGGL(lpp::attn_bwd_dkdv_kernel, dim3(kvblocks, B * HKV), dim3(512), lds,
+    const size_t lds = dbuf ? (131072 + 1024) : (65536 + 512);
+    auto kfn = dbuf ? lpp::attn_bwd_dkdv_kernel<true> : lpp::attn_bwd_dkdv_kernel<false>;
+    hipLaunchKernelGGL(kfn, dim3(kvblocks, B * HKV), dim3(512), lds,
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                        (const short*)v.data_ptr(), (const short*)dO.data_ptr(),
                        lse2.data_ptr<float>(), delta.data_ptr<float>(),
