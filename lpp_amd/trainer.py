@@ -181,6 +181,10 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
                     logger.info("eval @ step %d: loss %.4f", step, ev)
                     if wandb:
                         wandb.log({"eval_loss": ev}, step=step)
+            if cfg.logging_steps and step % cfg.logging_steps == 0:
+                # fold + reset device timers on EVERY rank (bounds the event
+                # backlog); only rank 0 logs its split
+                t = engine.timer_summary(reset=True)
             if rank0 and cfg.logging_steps and step % cfg.logging_steps == 0:
                 avg = tr_loss / cfg.logging_steps
                 tr_loss = 0.0
@@ -189,6 +193,11 @@ def train(cfg: TrainConfig, engine: PipelineEngine, resume_step: int = 0) -> dic
                     "loss": round(avg, 4),
                     "lr": engine.get_lr(),
                     "s/step": round(engine.last_step_time, 3),
+                    "fwd_s": round(t["forward"], 2),
+                    "bwd_s": round(t["backward"], 2),
+                    "p2p_s": round(t["p2p"], 2),
+                    "allreduce_s": round(t["allreduce"], 2),
+                    "optim_s": round(t["optimizer"], 2),
                 }
                 logger.info("%s", msg)
                 if wandb:
