@@ -206,6 +206,17 @@ class PipelineEngine:
             if cur:
                 self._push_bucket(cur, offs)
             self.optimizer.on_accumulate = self._on_grad_accumulated
+        # Initialize every communicator at a controlled point (RCCL comm
+        # creation is collective; doing it lazily inside the overlapped
+        # schedule would interleave group inits across ranks).
+        self.p2p.warmup_channels()
+        if dist.is_initialized():
+            warm = torch.zeros(1, device=device)
+            if grid.dp_degree > 1:
+                dist.all_reduce(warm, group=grid.dp_group)
+            if grid.num_stages > 1:
+                dist.all_reduce(warm, group=grid.pipe_group)
+
         try:
             wd_s = float(os.environ.get("LPP_WATCHDOG_S", "") or
                          getattr(config, "watchdog_timeout_s", 0) or 0)

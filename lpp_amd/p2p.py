@@ -71,6 +71,28 @@ class PipeP2P:
     def _empty(self) -> torch.Tensor:
         return torch.empty(self.act_shape, dtype=self.dtype, device=self.device)
 
+    def warmup_channels(self) -> None:
+        """One tiny ring exchange per channel so the RCCL communicators for
+        pipe_fwd/pipe_bwd initialize at a controlled point (engine init)
+        instead of lazily inside the overlapped 1F1B schedule — communicator
+        creation is collective per group, and first-use inside a schedule
+        interleaves two groups' inits across ranks."""
+        if not dist.is_initialized() or self.grid.num_stages <= 1:
+            return
+        t = torch.ones(1, dtype=self.dtype, device=self.device)
+        for send_rank, recv_rank, group in (
+            (self.grid.next_rank, self.grid.prev_rank, self._fwd_group),
+            (self.grid.prev_rank, self.grid.next_rank, self._bwd_group),
+        ):
+            reqs = []
+            if recv_rank is not None:
+                buf = torch.empty(1, dtype=self.dtype, device=self.device)
+                reqs.append(dist.irecv(buf, recv_rank, group=group))
+            if send_rank is not None:
+                reqs.append(dist.isend(t, send_rank, group=group))
+            for r in reqs:
+                r.wait()
+
     # -- async primitives (the engine's hot path) ---------------------------
     def isend_forward(self, tensor: torch.Tensor) -> Optional[Pending]:
         """Post the activation send to the next stage on the fwd channel."""
