@@ -234,11 +234,27 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
   ld_tile(0, 1, kreg1, vreg1);
   write_tile(0, 0, kreg0, vreg0);
   write_tile(0, 1, kreg1, vreg1);
+  if (DBUF && DQ_KVB < kv_end) {  // T14: pre-load tile 1 to registers
+    ld_tile(DQ_KVB, 0, kreg0, vreg0);
+    ld_tile(DQ_KVB, 1, kreg1, vreg1);
+  }
   __syncthreads();
 
   for (int kv0 = 0, cur = 0; kv0 < kv_end; kv0 += DQ_KVB, cur ^= 1) {
     const bool have_next = kv0 + DQ_KVB < kv_end;
-    if (have_next) {
+    if (DBUF) {
+      // T14 write-after-barrier: tile t+1 (in registers) written into the
+      // other buffer right after the barrier, overlapping this tile's
+      // MFMAs; tile t+2's loads re-issue immediately.
+      if (have_next) {
+        write_tile(cur ^ 1, 0, kreg0, vreg0);
+        write_tile(cur ^ 1, 1, kreg1, vreg1);
+        if (kv0 + 2 * DQ_KVB < kv_end) {
+          ld_tile(kv0 + 2 * DQ_KVB, 0, kreg0, vreg0);
+          ld_tile(kv0 + 2 * DQ_KVB, 1, kreg1, vreg1);
+        }
+      }
+    } else if (have_next) {
       ld_tile(kv0 + DQ_KVB, 0, kreg0, vreg0);
       ld_tile(kv0 + DQ_KVB, 1, kreg1, vreg1);
     }
@@ -297,10 +313,10 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dq_kernel(
       __builtin_amdgcn_s_setprio(0);
     }
 
-    if (have_next) {
-      if (!DBUF) __syncthreads();  // everyone done reading the only buffer
-      write_tile(DBUF ? (cur ^ 1) : 0, 0, kreg0, vreg0);
-      write_tile(DBUF ? (cur ^ 1) : 0, 1, kreg1, vreg1);
+    if (!DBUF && have_next) {
+      __syncthreads();  // everyone done reading the only buffer
+      write_tile(0, 0, kreg0, vreg0);
+      write_tile(0, 1, kreg1, vreg1);
     }
     __syncthreads();
   }
@@ -453,11 +469,25 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
     ld_tile(qstart, 1, qreg1, dreg1);
     write_tile(0, qstart, 0, qreg0, dreg0);
     write_tile(0, qstart, 1, qreg1, dreg1);
+    if (DBUF && qstart + KV_QT < S) {  // T14: pre-load tile 1 to registers
+      ld_tile(qstart + KV_QT, 0, qreg0, dreg0);
+      ld_tile(qstart + KV_QT, 1, qreg1, dreg1);
+    }
     __syncthreads();
 
     for (int qt0 = qstart, cur = 0; qt0 < S; qt0 += KV_QT, cur ^= 1) {
       const bool have_next = qt0 + KV_QT < S;
-      if (have_next) {
+      if (DBUF) {
+        // T14 write-after-barrier (see dq kernel)
+        if (have_next) {
+          write_tile(cur ^ 1, qt0 + KV_QT, 0, qreg0, dreg0);
+          write_tile(cur ^ 1, qt0 + KV_QT, 1, qreg1, dreg1);
+          if (qt0 + 2 * KV_QT < S) {
+            ld_tile(qt0 + 2 * KV_QT, 0, qreg0, dreg0);
+            ld_tile(qt0 + 2 * KV_QT, 1, qreg1, dreg1);
+          }
+        }
+      } else if (have_next) {
         ld_tile(qt0 + KV_QT, 0, qreg0, dreg0);
         ld_tile(qt0 + KV_QT, 1, qreg1, dreg1);
       }
@@ -525,10 +555,10 @@ __global__ __launch_bounds__(512, 2) void attn_bwd_dkdv_kernel(
         }
       }
 
-      if (have_next) {
-        if (!DBUF) __syncthreads();  // everyone done reading the only buffer
-        write_tile(DBUF ? (cur ^ 1) : 0, qt0 + KV_QT, 0, qreg0, dreg0);
-        write_tile(DBUF ? (cur ^ 1) : 0, qt0 + KV_QT, 1, qreg1, dreg1);
+      if (!DBUF && have_next) {
+        __syncthreads();  // everyone done reading the only buffer
+        write_tile(0, qt0 + KV_QT, 0, qreg0, dreg0);
+        write_tile(0, qt0 + KV_QT, 1, qreg1, dreg1);
       }
       __syncthreads();
     }
