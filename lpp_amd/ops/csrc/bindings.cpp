@@ -21,6 +21,7 @@ std::vector<at::Tensor> attention_bwd(at::Tensor dO, at::Tensor q, at::Tensor k,
 at::Tensor mfma_test_16x16x32(at::Tensor A, at::Tensor B);
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw);
 void wgrad_f32_accum_pre(at::Tensor xT, at::Tensor dyT, at::Tensor dw);
+void wgrad_bf16d_pre(at::Tensor xT, at::Tensor dyT, at::Tensor out);
 at::Tensor transpose2d(at::Tensor in);
 void accum_bf16_f32(at::Tensor dst, at::Tensor src);
 std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int64_t in,
@@ -54,6 +55,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_f32_accum_pre", &wgrad_f32_accum_pre,
         "dW_f32 += dyT @ xT^T, pre-transposed k-contiguous operands (hipBLASLt, beta=1)");
   m.def("transpose2d", &transpose2d, "bf16/fp16 [R,C] -> [C,R] LDS-tiled transpose");
+  m.def("wgrad_bf16d_pre", &wgrad_bf16d_pre,
+        "bf16-D pre-transposed wgrad GEMM into a reusable scratch (beta 0)");
   m.def("accum_bf16_f32", &accum_bf16_f32, "dst_f32 += src_bf16 (vectorised)");
   m.def("wgrad_tune", &wgrad_tune,
         "exhaustive hipBLASLt solution sweep for a wgrad shape -> [(index, ms, name)]",

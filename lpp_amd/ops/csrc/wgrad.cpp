@@ -83,6 +83,9 @@ static std::unordered_map<AlgoKey, CachedPlan, AlgoKeyHash> g_plans;
 // kind 1 (pre-transposed): A = xT rm[in,T], B = dyT rm[out,T] — both
 //   contiguous along the contraction dim, the same class as the forward
 //   GEMM (~1.33-1.67 PF/s); operands produced by ops/csrc/transpose.hip.
+// kind 2: kind-1 layouts with bf16 D, beta 0 (the production bf16-D wgrad
+//   GEMM; accumulation happens in accum.hip) — swept to compare against
+//   torch.matmul's own pick for the same problem.
 static CachedPlan make_problem(int64_t T, int64_t in, int64_t out, int kind) {
   CachedPlan plan;
   LPP_CHECK_BLASLT(hipblasLtMatmulDescCreate(&plan.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
@@ -105,15 +108,17 @@ static CachedPlan make_problem(int64_t T, int64_t in, int64_t out, int kind) {
     LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.a, HIP_R_16BF, T, in, T));
     LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.b, HIP_R_16BF, T, out, T));
   }
-  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(&plan.c, HIP_R_32F, in, out, in));
+  LPP_CHECK_BLASLT(hipblasLtMatrixLayoutCreate(
+      &plan.c, kind == 2 ? HIP_R_16BF : HIP_R_32F, in, out, in));
   return plan;
 }
 
 // Device-timed run of one candidate; returns best-of-reps milliseconds, or a
 // huge value if the algo fails at runtime.
 static float time_algo(CachedPlan& plan, hipblasLtMatmulAlgo_t& algo, void* ax, void* by,
-                       void* cw, void* ws, int reps, hipStream_t stream) {
-  const float a1 = 1.0f, b1 = 1.0f;
+                       void* cw, void* ws, int reps, hipStream_t stream,
+                       float beta = 1.0f) {
+  const float a1 = 1.0f, b1 = beta;
   hipEvent_t ev0, ev1;
   LPP_CHECK_HIP(hipEventCreate(&ev0));
   LPP_CHECK_HIP(hipEventCreate(&ev1));
@@ -161,13 +166,15 @@ static void pick_heuristic(CachedPlan& plan, int64_t T, int64_t in, int64_t out,
     auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
     auto sx = kind == 0 ? at::empty({T, in}, opt) : at::empty({in, T}, opt);
     auto sy = kind == 0 ? at::empty({T, out}, opt) : at::empty({out, T}, opt);
-    auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
+    auto sw = kind == 2 ? at::zeros({out, in}, opt)
+                        : at::zeros({out, in}, opt.dtype(at::kFloat));
     auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
     auto stream = current_stream();
     float best = 1e30f;
     for (int i = 0; i < found; ++i) {
       float ms = time_algo(plan, results[i].algo, sx.data_ptr(), sy.data_ptr(),
-                           sw.data_ptr(), wsbuf.data_ptr(), 1, stream);
+                           sw.data_ptr(), wsbuf.data_ptr(), 1, stream,
+                           kind == 2 ? 0.0f : 1.0f);
       if (ms < best) {
         best = ms;
         plan.algo = results[i].algo;
@@ -239,14 +246,17 @@ std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int6
   LPP_CHECK_BLASLT(hipblaslt_ext::getAllAlgos(
       lt_ctx().handle, hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
       kind == 0 ? HIPBLAS_OP_N : HIPBLAS_OP_T, kind == 0 ? HIPBLAS_OP_T : HIPBLAS_OP_N,
-      HIP_R_16BF, HIP_R_16BF, HIP_R_32F, HIP_R_32F, HIPBLAS_COMPUTE_32F, all));
+      HIP_R_16BF, HIP_R_16BF, kind == 2 ? HIP_R_16BF : HIP_R_32F,
+      kind == 2 ? HIP_R_16BF : HIP_R_32F, HIPBLAS_COMPUTE_32F, all));
   auto opt = at::TensorOptions().dtype(at::kBFloat16).device(at::kCUDA);
   auto sx = kind == 0 ? at::empty({T, in}, opt) : at::empty({in, T}, opt);
   auto sy = kind == 0 ? at::empty({T, out}, opt) : at::empty({out, T}, opt);
-  auto sw = at::zeros({out, in}, opt.dtype(at::kFloat));
+  auto sw = kind == 2 ? at::zeros({out, in}, opt)
+                      : at::zeros({out, in}, opt.dtype(at::kFloat));
   auto wsbuf = at::empty({(int64_t)kWorkspaceBytes}, opt.dtype(at::kByte));
   auto stream = current_stream();
-  const float a1 = 1.0f, b1 = 1.0f;
+  const float a1 = 1.0f;
+  const float b1 = kind == 2 ? 0.0f : 1.0f;
   std::vector<std::tuple<int64_t, double, std::string>> timed;
   for (auto& cand : all) {
     size_t ws_needed = 0;
@@ -255,7 +265,8 @@ std::vector<std::tuple<int64_t, double, std::string>> wgrad_tune(int64_t T, int6
         ws_needed);
     if (ok != HIPBLAS_STATUS_SUCCESS || ws_needed > kWorkspaceBytes) continue;
     float ms = time_algo(plan, cand.algo, sx.data_ptr(), sy.data_ptr(), sw.data_ptr(),
-                         wsbuf.data_ptr(), (int)reps, stream);
+                         wsbuf.data_ptr(), (int)reps, stream,
+                         kind == 2 ? 0.0f : 1.0f);
     if (ms >= 1e29f) continue;
     int idx = hipblaslt_ext::getIndexFromAlgo(cand.algo);
     timed.emplace_back(idx, (double)ms,
@@ -294,7 +305,31 @@ std::tuple<int64_t, std::string> wgrad_current_algo(int64_t T, int64_t in, int64
   return {idx, hipblaslt_ext::getKernelNameFromAlgo(lt_ctx().handle, plan.algo)};
 }
 
+// Production bf16-D pre-transposed wgrad GEMM (kind 2): out = dyT @ xT^T
+// into a caller-provided bf16 buffer (beta 0), tunable/pinnable like the
+// other kinds.
+void wgrad_bf16d_pre(at::Tensor xT, at::Tensor dyT, at::Tensor out) {
+  TORCH_CHECK(xT.is_cuda() && xT.scalar_type() == at::kBFloat16 && xT.dim() == 2);
+  TORCH_CHECK(dyT.is_cuda() && dyT.scalar_type() == at::kBFloat16 && dyT.dim() == 2);
+  TORCH_CHECK(out.is_cuda() && out.scalar_type() == at::kBFloat16 && out.is_contiguous());
+  TORCH_CHECK(xT.is_contiguous() && dyT.is_contiguous());
+  const int64_t in = xT.size(0), T = xT.size(1), outd = dyT.size(0);
+  TORCH_CHECK(dyT.size(1) == T && out.size(0) == outd && out.size(1) == in);
+  CachedPlan plan = get_plan(T, in, outd, 2);
+  auto workspace = at::empty({(int64_t)kWorkspaceBytes},
+                             xT.options().dtype(at::kByte));
+  const float alpha = 1.0f, beta = 0.0f;
+  LPP_CHECK_BLASLT(hipblasLtMatmul(
+      lt_ctx().handle, plan.op, &alpha, xT.data_ptr(), plan.a, dyT.data_ptr(), plan.b,
+      &beta, out.data_ptr(), plan.c, out.data_ptr(), plan.c, &plan.algo,
+      workspace.data_ptr(), kWorkspaceBytes, current_stream()));
+}
+
 }  // namespace lpp
+
+void wgrad_bf16d_pre(at::Tensor xT, at::Tensor dyT, at::Tensor out) {
+  lpp::wgrad_bf16d_pre(xT, dyT, out);
+}
 
 void wgrad_f32_accum(at::Tensor x, at::Tensor dy, at::Tensor dw) {
   lpp::wgrad_f32_accum(x, dy, dw);
