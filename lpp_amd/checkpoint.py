@@ -151,18 +151,86 @@ def load_engine_checkpoint(engine, ckpt_dir: str, tag: Optional[str] = None,
 
     client_state = None
     if not load_module_only:
-        f = Path(ckpt_dir) / tag / ENGINE_FMT.format(stage=grid.stage_id, dp=grid.dp_id)
+        step_dir = Path(ckpt_dir) / tag
+        f = step_dir / ENGINE_FMT.format(stage=grid.stage_id, dp=grid.dp_id)
+        es = None
         if f.exists():
             es = torch.load(f, map_location="cpu", weights_only=False)
+            if es["optimizer"].get("shard_world", 1) != engine.optimizer.shard_world:
+                es = None  # dp-degree changed: reshard below
+        if es is not None:
             engine.load_state_dict_local(es)
             client_state = es.get("client_state")
         else:
-            logger.warning(
-                "no engine state shard at %s — module-only load (converted checkpoint?)", f
-            )
+            old = sorted(step_dir.glob(
+                f"engine_state_pp{grid.stage_id:02d}_dp*.pt"))
+            if old:
+                client_state = _load_resharded(engine, old)
+            else:
+                logger.warning(
+                    "no engine state shard at %s — module-only load (converted "
+                    "checkpoint?)", f)
     if dist.is_initialized():
         dist.barrier()
     return client_state
+
+
+def _load_resharded(engine, shard_files) -> Optional[dict]:
+    """Resume optimizer state saved at a DIFFERENT dp_degree (ZeRO-1 shards
+    or replicated plain-DP state): reconstruct the full flat fp32 state from
+    every saved shard of this stage, then cut this engine's view of it —
+    plain per-param copies, or this rank's 1/dp shard (with new padding).
+    The reference stack cannot do this (DeepSpeed ZeRO resume is
+    world-size-pinned)."""
+    sds = [torch.load(f, map_location="cpu", weights_only=False) for f in shard_files]
+    base = sds[0]
+    opt_sds = [sd["optimizer"] for sd in sds]
+    old_world = opt_sds[0].get("shard_world", 1)
+    opt = engine.optimizer
+    total = sum(p.numel() for p in opt.params)
+    if old_world > 1 and len(sds) != old_world:
+        raise FileNotFoundError(
+            f"ZeRO-1 reshard needs all {old_world} saved shards of this stage, "
+            f"found {len(sds)}: {[str(f) for f in shard_files]}")
+    logger.info("resharding optimizer state: saved dp=%d -> engine dp shard_world=%d",
+                old_world, opt.shard_world)
+
+    def full_vec(key: str) -> torch.Tensor:
+        if old_world == 1:
+            parts = [t.reshape(-1).float() for t in opt_sds[0][key]]
+            return torch.cat(parts)[:total]
+        cat = torch.cat([osd[key][0].reshape(-1).float() for osd in opt_sds])
+        return cat[:total]  # strip the old dp padding
+
+    with torch.no_grad():
+        for key, dsts in (("masters", opt.masters), ("exp_avg", opt.exp_avg),
+                          ("exp_avg_sq", opt.exp_avg_sq)):
+            vec = full_vec(key)
+            if opt.is_sharded:
+                padded = torch.zeros(opt.padded_total, dtype=torch.float32)
+                padded[:total] = vec
+                dsts[0].copy_(padded[opt.shard_slice].to(dsts[0].device))
+            else:
+                off = 0
+                for p, dst in zip(opt.params, dsts):
+                    n = p.numel()
+                    dst.copy_(vec[off:off + n].view_as(p).to(dst.device))
+                    off += n
+        # re-sync model params from the (re)built masters
+        if opt.is_sharded:
+            opt.param_shard.copy_(opt.masters[0])
+            opt._maybe_allgather_params()
+        else:
+            for p, m in zip(opt.params, opt.masters):
+                p.data.copy_(m.to(p.dtype))
+    opt.step_count = opt_sds[0]["step_count"]
+    opt.lr = opt_sds[0].get("lr", opt.lr)
+    engine.lr_scheduler.load_state_dict(base["lr_scheduler"])
+    engine.global_steps = base.get("global_steps", 0)
+    engine.skipped_steps = base.get("skipped_steps", 0)
+    if engine.loss_scaler is not None and base.get("loss_scale"):
+        engine.loss_scaler.scale = base["loss_scale"]
+    return base.get("client_state")
 
 
 def parse_checkpoint_step(path: str) -> int:

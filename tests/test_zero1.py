@@ -138,3 +138,123 @@ def test_zero1_save_resume(tmp_path):
     for r in range(2):
         ref, got = res[r]
         assert ref == pytest.approx(got, rel=1e-5, abs=1e-6), (ref, got)
+
+
+# ---------------------------------------------------------------- reshard
+def _train_save(rank, world, zero_stage, outdir, steps=3):
+    from lpp_amd.checkpoint import save_engine_checkpoint
+
+    # reuse _train's engine construction by inlining a shortened version
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset
+    from lpp_amd.engine import PipelineEngine
+    from lpp_amd.models import get_layers_from_config, init_pipeline_weights, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+    import torch
+
+    mcfg = model_config("llama-tiny", num_layers=2, max_seq_len=64)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=64, dtype="fp32",
+                      zero_stage=zero_stage)
+    cfg.optimizer.lr = 1e-3
+    grid = ProcessGrid(world, rank, num_stages=1)
+    grid.build_groups()
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    init_pipeline_weights(module, mcfg, seed=5)
+    engine = PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    ds = SyntheticCausalLMDataset(64, 64, mcfg.vocab_size, seed=11)
+    loader = torch.utils.data.DataLoader(
+        ds, batch_size=2, shuffle=False, collate_fn=CausalLMCollator(64),
+        sampler=torch.utils.data.distributed.DistributedSampler(
+            ds, num_replicas=grid.dp_degree, rank=grid.dp_id, shuffle=False))
+    it = iter(RepeatingLoader(loader))
+    for _ in range(steps):
+        engine.train_batch(it)
+    save_engine_checkpoint(engine, outdir, tag="global_step3")
+    with torch.no_grad():
+        fp = torch.cat([p.reshape(-1).float() for p in module.parameters()]).clone()
+    return fp, engine.optimizer.step_count
+
+
+def _resume_single(zero_stage_new, outdir):
+    """Load the dp2 checkpoint into a WORLD-1 engine (dp degree changed)."""
+    from lpp_amd.checkpoint import load_engine_checkpoint
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.engine import PipelineEngine
+    from lpp_amd.models import get_layers_from_config, init_pipeline_weights, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+    import torch
+
+    mcfg = model_config("llama-tiny", num_layers=2, max_seq_len=64)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=64, dtype="fp32",
+                      zero_stage=zero_stage_new)
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    init_pipeline_weights(module, mcfg, seed=99)  # different init: load must win
+    engine = PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    load_engine_checkpoint(engine, outdir)
+    with torch.no_grad():
+        fp = torch.cat([p.reshape(-1).float() for p in module.parameters()]).clone()
+    masters = torch.cat([m.reshape(-1) for m in engine.optimizer.masters])
+    return fp, masters, engine.optimizer.step_count, engine.global_steps
+
+
+def test_zero1_resume_across_dp_degree(tmp_path):
+    """A ZeRO-1 dp=2 checkpoint resumes on dp=1 (sharded state regathered)."""
+    out = str(tmp_path / "z")
+    saved = run_dist(2, _train_save, 1, out)
+    fp_saved = saved[0][0]
+    fp, masters, step_count, gsteps = _resume_single(0, out)
+    assert torch.allclose(fp, fp_saved, atol=1e-6)
+    # masters must equal the fp32 params (fp32 run)
+    assert torch.allclose(masters[: fp.numel()], fp_saved, atol=1e-6)
+    assert step_count == saved[0][1]
+    assert gsteps == 3
+
+
+def test_plain_dp_resume_across_dp_degree(tmp_path):
+    """A plain-DP dp=2 checkpoint (replicated states) resumes on dp=1."""
+    out = str(tmp_path / "p")
+    saved = run_dist(2, _train_save, 0, out)
+    fp, masters, step_count, _ = _resume_single(0, out)
+    assert torch.allclose(fp, saved[0][0], atol=1e-6)
+    assert step_count == saved[0][1]
+
+
+def _resume_world(rank, world, outdir):
+    """Load the dp2 ZeRO-1 checkpoint into a dp=world sharded engine."""
+    from lpp_amd.checkpoint import load_engine_checkpoint
+    from lpp_amd.config import TrainConfig, model_config
+    from lpp_amd.engine import PipelineEngine
+    from lpp_amd.models import get_layers_from_config, init_pipeline_weights, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+    import torch
+
+    mcfg = model_config("llama-tiny", num_layers=2, max_seq_len=64)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=64, dtype="fp32",
+                      zero_stage=1)
+    grid = ProcessGrid(world, rank, 1)
+    grid.build_groups()
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=torch.device("cpu"), dtype=torch.float32)
+    init_pipeline_weights(module, mcfg, seed=77)
+    engine = PipelineEngine(module, cfg, grid, device=torch.device("cpu"))
+    load_engine_checkpoint(engine, outdir)
+    with torch.no_grad():
+        fp = torch.cat([p.reshape(-1).float() for p in module.parameters()]).clone()
+    return fp
+
+
+def test_zero1_resume_dp2_to_dp4(tmp_path):
+    out = str(tmp_path / "z24")
+    saved = run_dist(2, _train_save, 1, out)
+    got = run_dist(4, _resume_world, out)
+    for r in range(4):
+        assert torch.allclose(got[r], saved[0][0], atol=1e-6), r
