@@ -206,6 +206,12 @@ class PipelineEngine:
             if cur:
                 self._push_bucket(cur, offs)
             self.optimizer.on_accumulate = self._on_grad_accumulated
+            # The fused-wgrad linear path accumulates into main_grad
+            # directly (no p.grad, so the autograd post-accumulate hook
+            # never fires for those weights) — it notifies through this
+            # per-param stamp instead (ops/linear.py backward).
+            for p in self.optimizer.params:
+                p._on_accumulate = self._on_grad_accumulated
         # Initialize every communicator at a controlled point (RCCL comm
         # creation is collective; doing it lazily inside the overlapped
         # schedule would interleave group inits across ranks).
@@ -245,7 +251,10 @@ class PipelineEngine:
         collective rides the comm stream while backward keeps computing)."""
         if not self._grad_finalizing:
             return
-        b = self._buckets[self._bucket_of[id(p)]]
+        bi = self._bucket_of.get(id(p))
+        if bi is None:  # unknown tensor identity (re-wrapped save) — boundary
+            return      # fallback covers its bucket
+        b = self._buckets[bi]
         b["remaining"] -= 1
         if b["remaining"] == 0 and not b["launched"]:
             b["launched"] = True

@@ -127,6 +127,11 @@ class _LinearWgradF32(torch.autograd.Function):
                 ext.wgrad_f32_accum_pre(xT, dyT, ctx.main_grad)
         else:
             ext.wgrad_f32_accum(x2, dy2, ctx.main_grad)
+        # This path bypasses autograd's grad accumulation, so the engine's
+        # DP-bucket overlap (post-accumulate hooks) must be notified here.
+        cb = getattr(weight, "_on_accumulate", None)
+        if cb is not None:
+            cb(weight)
         return dx, None
 
 

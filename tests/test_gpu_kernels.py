@@ -361,3 +361,20 @@ def test_wgrad_bf16d_matches_f32d(monkeypatch):
     g32 = run("0")
     err = (g16 - g32).abs().max() / g32.abs().max().clamp(min=1e-3)
     assert err < 1e-2, err
+
+
+def test_fused_wgrad_notifies_accumulate():
+    """The fused wgrad path must fire the per-param accumulate notifier the
+    DP-bucket overlap relies on (no p.grad -> no autograd hook)."""
+    from lpp_amd.ops.linear import lp_linear
+
+    w = torch.nn.Parameter(
+        torch.randn(96, 128, device="cuda", dtype=torch.bfloat16) * 0.05)
+    w.main_grad = torch.zeros(96, 128, device="cuda", dtype=torch.float32)
+    fired = []
+    w._on_accumulate = lambda p: fired.append(p is w)
+    x = torch.randn(2, 64, 128, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    lp_linear(x, w).backward(torch.randn(2, 64, 96, device="cuda",
+                                         dtype=torch.bfloat16))
+    assert fired == [True]
