@@ -53,8 +53,13 @@ def choose_schedule(model_cfg, num_stages: int, layers_per_stage: int,
     apl = act_bytes_per_layer(seq_len, model_cfg.hidden_size,
                               model_cfg.intermediate_size, micro_batch_size)
     in_flight = min(num_stages, gas)  # stage 0 holds the most microbatches
-    stage_params = (model_cfg.num_params() // max(num_stages, 1)
-                    + 2 * model_cfg.vocab_size * model_cfg.hidden_size)
+    # Worst-stage parameter bytes: an even slice, plus the embedding OR the
+    # LM head for the edge stages when partitioned (num_params() already
+    # counts both, so a single-stage run adds nothing — double-counting
+    # them forced needless recompute at 128k-vocab models).
+    vocab_extra = (model_cfg.vocab_size * model_cfg.hidden_size
+                   if num_stages > 1 else 0)
+    stage_params = model_cfg.num_params() // max(num_stages, 1) + vocab_extra
     free_layers = max(0, int((budget - stage_params * BYTES_PER_PARAM)
                              // (apl * in_flight)))
     ckpt = max(0, layers_per_stage - free_layers)
