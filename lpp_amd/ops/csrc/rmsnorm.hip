@@ -87,10 +87,19 @@ __global__ void rmsnorm_bwd_kernel_v(const T* __restrict__ dy, const T* __restri
   // N_IT is compile-time so the packet/partial arrays stay in registers
   // (a runtime bound would dynamic-index them into scratch).
   alignas(16) float dwacc[N_IT][VEC];
+  // w is row-invariant but the block-reduce barrier stops the compiler
+  // hoisting its loads out of the row loop — preload to registers once.
+  alignas(16) float wreg[N_IT][VEC];
 #pragma unroll
-  for (int it = 0; it < N_IT; ++it)
+  for (int it = 0; it < N_IT; ++it) {
+    const int i = threadIdx.x * VEC + it * BLOCK * VEC;
+    *reinterpret_cast<Pack<float, 4>*>(&wreg[it][0]) =
+        *reinterpret_cast<const Pack<float, 4>*>(w + i);
+    *reinterpret_cast<Pack<float, 4>*>(&wreg[it][4]) =
+        *reinterpret_cast<const Pack<float, 4>*>(w + i + 4);
 #pragma unroll
     for (int v = 0; v < VEC; ++v) dwacc[it][v] = 0.f;
+  }
 
   for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const T* dyr = dy + row * H;
@@ -106,7 +115,7 @@ __global__ void rmsnorm_bwd_kernel_v(const T* __restrict__ dy, const T* __restri
       xbuf[it] = *reinterpret_cast<const PV*>(xr + i);
 #pragma unroll
       for (int v = 0; v < VEC; ++v)
-        dot += to_f32(dybuf[it].v[v]) * w[i + v] * to_f32(xbuf[it].v[v]);
+        dot += to_f32(dybuf[it].v[v]) * wreg[it][v] * to_f32(xbuf[it].v[v]);
     }
     const float total = block_reduce_sum<BLOCK>(dot, red);
     const float k = total * inv * inv * inv / (float)H;
@@ -118,7 +127,7 @@ __global__ void rmsnorm_bwd_kernel_v(const T* __restrict__ dy, const T* __restri
       for (int v = 0; v < VEC; ++v) {
         const float d = to_f32(dybuf[it].v[v]);
         const float xi = to_f32(xbuf[it].v[v]);
-        out.v[v] = from_f32<T>(d * w[i + v] * inv - xi * k);
+        out.v[v] = from_f32<T>(d * wreg[it][v] * inv - xi * k);
         dwacc[it][v] += d * xi * inv;
       }
       *reinterpret_cast<PV*>(dxr + i) = out;
