@@ -39,7 +39,9 @@ class MixedPrecisionAdamW:
     against the (reduce-scattered) gradient shard, and all-gathers the
     updated flat parameters.  Memory for optimizer state drops from
     12 B/param to 12/dp B/param per rank.  Checkpoints of the optimizer
-    state are shard-local: resume requires the same dp_degree."""
+    state are shard-local; resuming at a DIFFERENT dp_degree goes through
+    checkpoint._load_resharded (full-state regather + re-cut), while
+    loading a mismatched shard directly raises."""
 
     def __init__(
         self,

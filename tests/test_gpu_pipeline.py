@@ -116,3 +116,38 @@ def test_pipeline_generate_single_stage_gpu():
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     print(f"\ndecode: {32 / dt:.1f} tok/s on 4/80 of a 65B (B=1, prefill 128)")
+
+
+def test_single_gpu_fp16_loss_scaled_training():
+    """Reference-parity fp16 regime on GPU: dynamic loss scaler + fused
+    AdamW; finite decreasing loss, no scaler collapse."""
+    from lpp_amd.config import TrainConfig, model_config, torch_dtype
+    from lpp_amd.data import CausalLMCollator, RepeatingLoader, SyntheticCausalLMDataset
+    from lpp_amd.engine import PipelineEngine
+    from lpp_amd.models import get_layers_from_config, loss_fn
+    from lpp_amd.pipeline_module import PipelineModule
+    from lpp_amd.topology import ProcessGrid
+
+    device = torch.device("cuda", 0)
+    mcfg = model_config("llama-7b", num_layers=2, max_seq_len=128, vocab_size=32000)
+    cfg = TrainConfig(model=mcfg, num_stages=1, micro_batch_size=2,
+                      gradient_accumulation_steps=2, seq_len=128, dtype="fp16")
+    cfg.optimizer.lr = 1e-4
+    grid = ProcessGrid(1, 0, 1)
+    module = PipelineModule(get_layers_from_config(mcfg), grid, loss_fn=loss_fn,
+                            device=device, dtype=torch_dtype("fp16"))
+    with torch.no_grad():
+        for p in module.parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, 0.02)
+    engine = PipelineEngine(module, cfg, grid, device=device)
+    assert engine.loss_scaler is not None
+    ds = SyntheticCausalLMDataset(4, 128, mcfg.vocab_size, seed=3)
+    loader = torch.utils.data.DataLoader(ds, batch_size=2, shuffle=False,
+                                         collate_fn=CausalLMCollator(128))
+    it = iter(RepeatingLoader(loader))
+    losses = [float(engine.train_batch(it)) for _ in range(5)]
+    assert all(l == l for l in losses), losses
+    assert engine.loss_scaler.scale >= 1.0
+    assert engine.global_steps == 5
+    assert losses[-1] < losses[0] + 0.5, losses
