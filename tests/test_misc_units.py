@@ -98,3 +98,42 @@ def test_step_timer():
     time.sleep(0.01)
     dt = t.stop()
     assert dt >= 0.01 and t.mean >= 0.01
+
+
+def test_device_timers_cpu_path():
+    from lpp_amd.utils.timers import DeviceTimers
+    import time as _t
+
+    t = DeviceTimers(None)
+    with t.section("a"):
+        _t.sleep(0.01)
+    with t.section("a"):
+        pass
+    with t.section("b"):
+        pass
+    out = t.summary(reset=True)
+    assert out["a"] >= 0.01
+    assert "b" in out
+    assert t.totals_nosync() == {}
+
+
+def _rz_body(rank, world):
+    import time as _t
+    from lpp_amd.utils import rank_zero_first
+
+    with rank_zero_first(rank):
+        t_enter = _t.time()
+        if rank == 0:
+            _t.sleep(0.3)
+        t_exit = _t.time()
+    return t_enter, t_exit
+
+
+def test_rank_zero_first_ordering():
+    """Rank 0 must complete the body before any other rank enters it."""
+    from tests.dist_utils import run_dist
+
+    got = run_dist(2, _rz_body)
+    rank0_exit = got[0][1]
+    rank1_enter = got[1][0]
+    assert rank1_enter >= rank0_exit - 0.05, (got,)

@@ -124,3 +124,13 @@ def test_pp8_matches_single_process():
     got = run_dist(8, run_steps, 8, 2, 8, timeout=600.0)
     for a, b in zip(base, got[0]):
         assert abs(a - b) < 1e-3, (base, got[0])
+
+
+def test_pp3_matches_single_process():
+    """Odd stage count: stage 1 is a pure middle stage under the overlapped
+    schedule (pre-posted recvs on both channels simultaneously)."""
+    base = _single_process_baseline(steps=2, gas=6)
+    got = run_dist(3, run_steps, 3, 2, 6)
+    for r in range(3):
+        for a, b in zip(base, got[r]):
+            assert abs(a - b) < 1e-3, (base, got[r])
