@@ -134,3 +134,15 @@ def test_pp3_matches_single_process():
     for r in range(3):
         for a, b in zip(base, got[r]):
             assert abs(a - b) < 1e-3, (base, got[r])
+
+
+@pytest.mark.slow
+@pytest.mark.parametrize("world,gas", [(3, 2), (3, 4), (5, 3), (5, 7), (6, 6)])
+def test_schedule_property_sweep(world, gas):
+    """Randomized-shape sweep of the overlapped 1F1B schedule: every
+    (depth, microbatch-count) combination — including gas < depth and
+    non-power-of-two depths — must reproduce the monolithic trajectory."""
+    base = _single_process_baseline(steps=2, gas=gas)
+    got = run_dist(world, run_steps, world, 2, gas, timeout=420.0)
+    for a, b in zip(base, got[0]):
+        assert abs(a - b) < 1e-3, (world, gas, base, got[0])
