@@ -284,6 +284,54 @@ class TextCollator:
         return {"input_ids": ids, "labels": labels}
 
 
+def combine_on_length(a: torch.Tensor, b: torch.Tensor,
+                      pad_value: int = 0) -> torch.Tensor:
+    """Stack two batches of id rows whose sequence lengths may differ,
+    padding the shorter to the longer (reference combine_tensor_on_length,
+    data/flan.py:173-178)."""
+    L = max(a.size(-1), b.size(-1))
+
+    def _pad(t: torch.Tensor) -> torch.Tensor:
+        if t.size(-1) == L:
+            return t
+        pad = torch.full((*t.shape[:-1], L - t.size(-1)), pad_value, dtype=t.dtype)
+        return torch.cat([t, pad], dim=-1)
+
+    return torch.cat([_pad(a), _pad(b)], dim=0)
+
+
+class WrappingCollator:
+    """Multi-task batching: wrap ANOTHER collator and merge this collator's
+    converted examples into its batch under ``<prefix>_*`` keys (reference
+    FlanCollatorOverCollator's wrapping mode, data/flan.py:263-309 — the
+    items are dicts carrying both the inner collator's example and a
+    ``field`` entry with this side's {"inputs","targets"} example, as
+    RoundRobinMixDataset produces).  Tensor keys present on both sides are
+    ALSO merged into one batch via length-padding (combine_on_length) so a
+    single forward can span both tasks."""
+
+    def __init__(self, inner_collator, tokenizer, max_seq_length: int,
+                 field: str = "flan", prefix: Optional[str] = None,
+                 merge_keys: Sequence[str] = ()):
+        self.inner = inner_collator
+        self.own = TextCollator(tokenizer, max_seq_length, field=None)
+        self.field = field
+        self.prefix = prefix if prefix is not None else field
+        self.merge_keys = tuple(merge_keys)
+
+    def __call__(self, batch: List[Dict]) -> Dict[str, torch.Tensor]:
+        inner_items = [{k: v for k, v in b.items() if k != self.field} for b in batch]
+        own_items = [b[self.field] for b in batch]
+        out = dict(self.inner(inner_items))
+        own = self.own(own_items)
+        for k, v in own.items():
+            out[f"{self.prefix}_{k}"] = v
+        for k in self.merge_keys:
+            pad = IGNORE_INDEX if k == "labels" else self.own.convert.tokenizer.pad_token_id
+            out[k] = combine_on_length(out[k], own[k], pad_value=pad)
+        return out
+
+
 class SimpleTokenizer:
     """Minimal offline whitespace tokenizer implementing the subset of the
     HF tokenizer protocol the data layer uses.  Exists because this

@@ -178,3 +178,49 @@ def test_text_collator_with_real_hf_tokenizer():
     # prompt region masked
     p0 = len(tok("the cat sat")["input_ids"])
     assert (out["labels"][0, :p0] == IGNORE_INDEX).all()
+
+
+def test_combine_on_length():
+    from lpp_amd.data import combine_on_length
+
+    a = torch.ones(2, 3, dtype=torch.long)
+    b = torch.full((1, 5), 2, dtype=torch.long)
+    out = combine_on_length(a, b, pad_value=9)
+    assert out.shape == (3, 5)
+    assert out[0].tolist() == [1, 1, 1, 9, 9]
+    assert out[2].tolist() == [2, 2, 2, 2, 2]
+
+
+def test_wrapping_collator_multi_task():
+    """Reference FlanCollatorOverCollator wrapping mode: inner batch +
+    flan_* keys + optional merged multi-task batch (data/flan.py:263-309)."""
+    from lpp_amd.data import (RoundRobinMixDataset, SimpleTokenizer, TextCollator,
+                              WrappingCollator)
+
+    tok = SimpleTokenizer(512)
+    main = [{"inputs": f"main q {i}", "targets": f"main a {i}"} for i in range(4)]
+    flan = [{"inputs": f"flan q {i}", "targets": f"flan a {i}"} for i in range(2)]
+    mix = RoundRobinMixDataset(main=_ListDS(main), flan=_ListDS(flan))
+    inner = TextCollator(tok, max_seq_length=16, field="main", pad_to_max=False)
+    coll = WrappingCollator(inner, tok, 16, field="flan",
+                            merge_keys=("input_ids", "labels"))
+    batch = [mix[i] for i in range(3)]
+    out = coll(batch)
+    assert "input_ids" in out and "labels" in out
+    assert "flan_input_ids" in out and "flan_labels" in out
+    # merged batch = inner rows + flan rows, padded to a common length
+    assert out["input_ids"].shape[0] == 6
+    assert out["input_ids"].shape[1] == out["labels"].shape[1]
+    # flan examples wrap around (index % len)
+    assert out["flan_input_ids"].shape[0] == 3
+
+
+class _ListDS(torch.utils.data.Dataset):
+    def __init__(self, items):
+        self.items = items
+
+    def __len__(self):
+        return len(self.items)
+
+    def __getitem__(self, i):
+        return self.items[i]
