@@ -63,6 +63,91 @@ class KVCache:
         return self.length
 
 
+class SlotKVCache:
+    """Row-view of ONE slot of a RaggedKVCache, exposing the KVCache API so
+    a single request's prefill runs through the unchanged flash path."""
+
+    def __init__(self, parent: "RaggedKVCache", slot: int):
+        self.parent = parent
+        self.slot = slot
+
+    @property
+    def k(self):
+        return self.parent.k[self.slot : self.slot + 1]
+
+    @property
+    def v(self):
+        return self.parent.v[self.slot : self.slot + 1]
+
+    @property
+    def length(self) -> int:
+        return int(self.parent.lengths[self.slot])
+
+    def append(self, k: torch.Tensor, v: torch.Tensor) -> int:
+        S = k.shape[1]
+        ln = self.length
+        self.parent.k[self.slot, ln : ln + S] = k[0]
+        self.parent.v[self.slot, ln : ln + S] = v[0]
+        self.parent.lengths[self.slot] = ln + S
+        return ln + S
+
+
+class RaggedKVCache:
+    """Slot-based KV cache with PER-ROW lengths — the substrate for
+    continuous batching (lpp_amd/serving.py): each slot holds an
+    independent request at its own decode position.  ``per_row`` marks the
+    ragged decode contract for LlamaAttention (per-row RoPE offsets +
+    key-validity mask).  Beyond the reference (training-only template)."""
+
+    per_row = True
+
+    def __init__(self, slots: int, max_len: int, kv_heads: int, head_dim: int,
+                 device, dtype):
+        self.k = torch.zeros(slots, max_len, kv_heads, head_dim, device=device, dtype=dtype)
+        self.v = torch.zeros_like(self.k)
+        self.lengths = torch.zeros(slots, dtype=torch.long, device=device)
+
+    def slot_view(self, slot: int) -> SlotKVCache:
+        return SlotKVCache(self, slot)
+
+    def free(self, slot: int) -> None:
+        self.lengths[slot] = 0
+
+
+class GatherKVCache:
+    """Decode-time view over the ACTIVE slots of a RaggedKVCache (ragged
+    per-row lengths).  ``append`` scatters each row's new K/V at that
+    row's own position."""
+
+    per_row = True
+
+    def __init__(self, parent: RaggedKVCache, slots: torch.Tensor):
+        self.parent = parent
+        self.slots = slots  # long tensor of active slot indices
+
+    @property
+    def k(self):
+        return self.parent.k[self.slots]
+
+    @property
+    def v(self):
+        return self.parent.v[self.slots]
+
+    @property
+    def lengths(self) -> torch.Tensor:
+        return self.parent.lengths[self.slots]
+
+    def append_one(self, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+        """Scatter single-token K/V ([N,1,Hkv,D]) at each active row's own
+        length; returns the new per-row lengths."""
+        lens = self.parent.lengths[self.slots]
+        self.parent.k[self.slots, lens] = k[:, 0]
+        self.parent.v[self.slots, lens] = v[:, 0]
+        new = lens + 1
+        self.parent.lengths[self.slots] = new
+        return new
+
+
 class LlamaAttention(nn.Module):
     """Self-attention with RoPE; q/k/v/o projections named for HF key parity."""
 
@@ -87,6 +172,31 @@ class LlamaAttention(nn.Module):
         cos, sin = ops.build_rope_cache(
             self.max_seq_len, self.head_dim, self.rope_theta, x.device
         )
+        if cache is not None and getattr(cache, "per_row", False):
+            # ragged single-token decode (continuous batching): every row
+            # sits at its OWN position; RoPE phases gathered per row and
+            # attention masked by per-row key validity.
+            assert S == 1, "ragged cache is a single-token decode contract"
+            lens = cache.lengths  # positions BEFORE append
+            q = ops.apply_rope_positions(q, cos, sin, lens)
+            k = ops.apply_rope_positions(k, cos, sin, lens)
+            new_lens = cache.append_one(k, v)
+            T = int(new_lens.max())
+            kc = cache.k[:, :T]
+            vc = cache.v[:, :T]
+            rep = self.num_heads // self.num_kv_heads
+            qt = q.transpose(1, 2)
+            kt = kc.transpose(1, 2)
+            vt = vc.transpose(1, 2)
+            if rep > 1:
+                kt = kt.repeat_interleave(rep, dim=1)
+                vt = vt.repeat_interleave(rep, dim=1)
+            mask = (torch.arange(T, device=x.device)[None, :]
+                    < new_lens[:, None])[:, None, None, :]
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qt, kt, vt, attn_mask=mask
+            ).transpose(1, 2).contiguous()
+            return self.o_proj(o.reshape(B, S, self.num_heads * self.head_dim))
         pos = cache.length if cache is not None else 0
         q = ops.apply_rope(q, cos, sin, pos_offset=pos)
         k = ops.apply_rope(k, cos, sin, pos_offset=pos)

@@ -94,7 +94,7 @@ def use_hip(*tensors: torch.Tensor) -> bool:
 
 
 from .rmsnorm import rmsnorm  # noqa: E402
-from .rope import build_rope_cache, apply_rope  # noqa: E402
+from .rope import build_rope_cache, apply_rope, apply_rope_positions  # noqa: E402
 from .swiglu import swiglu  # noqa: E402
 from .cross_entropy import shifted_cross_entropy  # noqa: E402
 from .attention import causal_attention  # noqa: E402
@@ -103,6 +103,7 @@ __all__ = [
     "rmsnorm",
     "build_rope_cache",
     "apply_rope",
+    "apply_rope_positions",
     "swiglu",
     "shifted_cross_entropy",
     "causal_attention",

@@ -78,3 +78,19 @@ def apply_rope(
     if use_hip(x):
         return _RopeHIP.apply(x.contiguous(), cos, sin, pos_offset)
     return apply_rope_ref(x, cos, sin, pos_offset)
+
+
+@torch.no_grad()
+def apply_rope_positions(
+    x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, positions: torch.Tensor
+) -> torch.Tensor:
+    """Ragged variant for continuous-batching decode: x [N, 1, H, D] with a
+    DIFFERENT position per row (positions [N] long).  Eager on purpose —
+    single-token decode is tiny and latency-bound."""
+    N, S, H, D = x.shape
+    assert S == 1, "positions variant is a single-token decode contract"
+    c = cos[positions].view(N, 1, 1, D // 2).to(torch.float32)
+    s = sin[positions].view(N, 1, 1, D // 2).to(torch.float32)
+    xf = x.float()
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
+    return torch.cat((x1 * c - x2 * s, x2 * c + x1 * s), dim=-1).to(x.dtype)
