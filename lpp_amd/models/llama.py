@@ -114,6 +114,25 @@ class RaggedKVCache:
         self.lengths[slot] = 0
 
 
+class MultiSlotKVCache:
+    """Batched-prefill view over several EMPTY slots of a RaggedKVCache:
+    the G admitted prompts (equal length) prefill as one [G, S] forward
+    through the unchanged flash path, K/V scattered into their rows."""
+
+    def __init__(self, parent: RaggedKVCache, slots: torch.Tensor):
+        self.parent = parent
+        self.slots = slots
+        self.length = 0  # uniform across the (empty) slots
+
+    def append(self, k: torch.Tensor, v: torch.Tensor) -> int:
+        S = k.shape[1]
+        self.parent.k[self.slots, self.length : self.length + S] = k
+        self.parent.v[self.slots, self.length : self.length + S] = v
+        self.length += S
+        self.parent.lengths[self.slots] = self.length
+        return self.length
+
+
 class GatherKVCache:
     """Decode-time view over the ACTIVE slots of a RaggedKVCache (ragged
     per-row lengths).  ``append`` scatters each row's new K/V at that
@@ -132,6 +151,12 @@ class GatherKVCache:
     @property
     def v(self):
         return self.parent.v[self.slots]
+
+    def kv_to(self, T: int):
+        """Gather the active rows narrowed to the first T positions in one
+        indexing op — the full-row gather copies slots x max_len per layer
+        per decode tick, which dominates decode time."""
+        return self.parent.k[self.slots, :T], self.parent.v[self.slots, :T]
 
     @property
     def lengths(self) -> torch.Tensor:
@@ -182,8 +207,10 @@ class LlamaAttention(nn.Module):
             k = ops.apply_rope_positions(k, cos, sin, lens)
             new_lens = cache.append_one(k, v)
             T = int(new_lens.max())
-            kc = cache.k[:, :T]
-            vc = cache.v[:, :T]
+            if hasattr(cache, "kv_to"):
+                kc, vc = cache.kv_to(T)
+            else:
+                kc, vc = cache.k[:, :T], cache.v[:, :T]
             rep = self.num_heads // self.num_kv_heads
             qt = q.transpose(1, 2)
             kt = kc.transpose(1, 2)
@@ -204,12 +231,13 @@ class LlamaAttention(nn.Module):
             o = ops.causal_attention(q, k, v)  # [B,S,H,D]
         else:
             total = cache.append(k, v)
-            kc = cache.k[:, :total]
-            vc = cache.v[:, :total]
             if S == total:
-                # prefill: plain causal over the whole prefix (flash path)
+                # prefill: plain causal over the whole prefix (flash path;
+                # no cache read-back — MultiSlotKVCache is write-only)
                 o = ops.causal_attention(q, k, v)
             else:
+                kc = cache.k[:, :total]
+                vc = cache.v[:, :total]
                 # decode: q attends the full cached prefix.  rows see
                 # positions <= their own: causal offset mask for S > 1,
                 # no mask needed for single-token decode.

@@ -31,7 +31,8 @@ from typing import Dict, List, Optional
 
 import torch
 
-from .models.llama import DecoderLayerPipe, RaggedKVCache, GatherKVCache
+from .models.llama import (DecoderLayerPipe, GatherKVCache, MultiSlotKVCache,
+                           RaggedKVCache)
 
 
 @dataclass
@@ -96,36 +97,58 @@ class ContinuousBatchingEngine:
 
     @torch.no_grad()
     def _admit(self) -> None:
-        """Move waiting requests into free slots; prefill each one."""
+        """Move waiting requests into free slots; requests with EQUAL
+        prompt lengths prefill together as one batched flash forward."""
         while self.waiting and self.free_slots:
-            req = self.waiting.popleft()
-            slot = self.free_slots.popleft()
-            req.slot = slot
+            group = [self.waiting.popleft()]
+            S0 = group[0].prompt.numel()
+            while (self.waiting and len(group) < len(self.free_slots)
+                   and self.waiting[0].prompt.numel() == S0):
+                group.append(self.waiting.popleft())
+            slots = [self.free_slots.popleft() for _ in group]
             for c in self.caches:
-                c.free(slot)
-            views = [c.slot_view(slot) for c in self.caches]
-            ids = req.prompt.view(1, -1).to(self.device)
-            logits = self._run_layers(ids, views)[0, -1]
-            self._last_logits[slot] = logits
-            self.active[slot] = req
+                for s in slots:
+                    c.free(s)
+            slot_t = torch.tensor(slots, dtype=torch.long, device=self.device)
+            views = [MultiSlotKVCache(c, slot_t) for c in self.caches]
+            ids = torch.stack([r.prompt for r in group]).to(self.device)
+            logits = self._run_layers(ids, views)[:, -1]
+            for i, (req, slot) in enumerate(zip(group, slots)):
+                req.slot = slot
+                self._last_logits[slot] = logits[i]
+                self.active[slot] = req
 
-    def _sample(self, req: Request, logits: torch.Tensor) -> int:
-        if req.temperature > 0:
-            probs = torch.softmax(logits.float() / req.temperature, dim=-1)
-            return int(torch.multinomial(probs, 1, generator=self.generator))
-        return int(logits.argmax())
+    def _sample_batched(self, slots: List[int]) -> List[int]:
+        """One device->host sync per tick (per-slot .item() calls dominated
+        the first version's decode time)."""
+        logits = torch.stack([self._last_logits[s] for s in slots])  # [N, V]
+        temps = torch.tensor([self.active[s].temperature for s in slots],
+                             device=logits.device)
+        toks = logits.argmax(dim=-1)
+        if bool((temps > 0).any()):
+            probs = torch.softmax(
+                logits.float() / temps.clamp(min=1e-6)[:, None], dim=-1)
+            gen = self.generator if (self.generator is not None and
+                                     str(self.generator.device) ==
+                                     str(logits.device)) else None
+            sampled = torch.multinomial(probs, 1, generator=gen).squeeze(-1)
+            toks = torch.where(temps > 0, sampled, toks)
+        return toks.tolist()
 
     @torch.no_grad()
     def step(self) -> List[str]:
         """One engine tick: admit -> sample each active slot's pending
-        logits -> retire finished -> ONE batched ragged decode for the
-        rest.  Returns the uids finished this tick."""
+        logits (batched, one sync) -> retire finished -> ONE batched
+        ragged decode for the rest.  Returns the uids finished this tick."""
         self._admit()
         done: List[str] = []
+        slots = sorted(self.active)
+        if not slots:
+            return done
         # sample from the logits produced by the previous forward (the
         # prefill for newly admitted requests)
-        for slot, req in list(self.active.items()):
-            tok = self._sample(req, self._last_logits[slot])
+        for slot, tok in zip(slots, self._sample_batched(slots)):
+            req = self.active[slot]
             req.generated.append(tok)
             if ((self.eos_token_id is not None and tok == self.eos_token_id)
                     or len(req.generated) >= req.max_new_tokens):
