@@ -136,13 +136,18 @@ class MultiSlotKVCache:
 class GatherKVCache:
     """Decode-time view over the ACTIVE slots of a RaggedKVCache (ragged
     per-row lengths).  ``append`` scatters each row's new K/V at that
-    row's own position."""
+    row's own position.  When the active slots form a contiguous range
+    (the common steady state — admission fills the lowest holes), pass
+    ``contiguous_range`` so ``kv_to`` is a zero-copy slice instead of an
+    advanced-indexing gather (the gather dominated decode tick time)."""
 
     per_row = True
 
-    def __init__(self, parent: RaggedKVCache, slots: torch.Tensor):
+    def __init__(self, parent: RaggedKVCache, slots: torch.Tensor,
+                 contiguous_range: Optional[tuple] = None):
         self.parent = parent
         self.slots = slots  # long tensor of active slot indices
+        self.range = contiguous_range  # (lo, hi) half-open, or None
 
     @property
     def k(self):
@@ -153,23 +158,33 @@ class GatherKVCache:
         return self.parent.v[self.slots]
 
     def kv_to(self, T: int):
-        """Gather the active rows narrowed to the first T positions in one
-        indexing op — the full-row gather copies slots x max_len per layer
-        per decode tick, which dominates decode time."""
+        """Active rows narrowed to the first T positions: a zero-copy slice
+        for a contiguous slot range, else one advanced-indexing gather."""
+        if self.range is not None:
+            lo, hi = self.range
+            return self.parent.k[lo:hi, :T], self.parent.v[lo:hi, :T]
         return self.parent.k[self.slots, :T], self.parent.v[self.slots, :T]
 
     @property
     def lengths(self) -> torch.Tensor:
+        if self.range is not None:
+            lo, hi = self.range
+            return self.parent.lengths[lo:hi]
         return self.parent.lengths[self.slots]
 
     def append_one(self, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
         """Scatter single-token K/V ([N,1,Hkv,D]) at each active row's own
         length; returns the new per-row lengths."""
-        lens = self.parent.lengths[self.slots]
-        self.parent.k[self.slots, lens] = k[:, 0]
-        self.parent.v[self.slots, lens] = v[:, 0]
+        lens = self.lengths
+        if self.range is not None:
+            lo, hi = self.range
+            rows = torch.arange(lo, hi, device=lens.device)
+        else:
+            rows = self.slots
+        self.parent.k[rows, lens] = k[:, 0]
+        self.parent.v[rows, lens] = v[:, 0]
         new = lens + 1
-        self.parent.lengths[self.slots] = new
+        self.parent.lengths[rows] = new
         return new
 
 

@@ -164,13 +164,20 @@ class ContinuousBatchingEngine:
                 self.free_slots.append(slot)
         if not self.active:
             return done
-        # one batched ragged decode over every active slot
+        # one batched ragged decode over every active slot (views memoized
+        # per active-set; contiguous slot ranges decode zero-copy)
         slots = sorted(self.active)
-        slot_t = torch.tensor(slots, dtype=torch.long, device=self.device)
+        key = tuple(slots)
+        if getattr(self, "_views_key", None) != key:
+            slot_t = torch.tensor(slots, dtype=torch.long, device=self.device)
+            rng = ((slots[0], slots[-1] + 1)
+                   if slots == list(range(slots[0], slots[-1] + 1)) else None)
+            self._views = [GatherKVCache(c, slot_t, contiguous_range=rng)
+                           for c in self.caches]
+            self._views_key = key
         toks = torch.tensor([self.active[s].generated[-1] for s in slots],
                             dtype=torch.long, device=self.device).view(-1, 1)
-        views = [GatherKVCache(c, slot_t) for c in self.caches]
-        logits = self._run_layers(toks, views)[:, -1]
+        logits = self._run_layers(toks, self._views)[:, -1]
         for i, s in enumerate(slots):
             self._last_logits[s] = logits[i]
         return done

@@ -46,34 +46,50 @@ def main() -> int:
                for _ in range(args.requests)]
     total_tokens = sum(lens)
 
-    # ---- continuous batching
+    # ---- continuous batching: record per-request completion times
     eng = ContinuousBatchingEngine(m, max_slots=args.slots, max_seq_len=args.max_seq)
     for i, (p, n) in enumerate(zip(prompts, lens)):
         eng.submit(Request(f"r{i}", p, n))
+    finish_c = {}
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     while eng.pending():
-        eng.step()
+        for uid in eng.step():
+            torch.cuda.synchronize()
+            finish_c[uid] = time.perf_counter() - t0
     torch.cuda.synchronize()
     t_cont = time.perf_counter() - t0
+    lat_c = sum(finish_c.values()) / len(finish_c)
 
     # ---- static batching: groups of `slots`, every request padded to the
-    # group's longest generation
+    # group's longest generation; a request completes when its GROUP does
+    finish_s = {}
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(0, args.requests, args.slots):
         grp = list(range(i, min(i + args.slots, args.requests)))
         ids = torch.stack([prompts[j] for j in grp])
         m.generate(ids, max_new_tokens=max(lens[j] for j in grp))
-    torch.cuda.synchronize()
+        torch.cuda.synchronize()
+        tg = time.perf_counter() - t0
+        for j in grp:
+            finish_s[j] = tg
     t_stat = time.perf_counter() - t0
+    lat_s = sum(finish_s.values()) / len(finish_s)
+    padded = sum(max(lens[j] for j in range(i, min(i + args.slots, args.requests)))
+                 * min(args.slots, args.requests - i)
+                 for i in range(0, args.requests, args.slots)) - total_tokens
 
     print(f"requests={args.requests} slots={args.slots} prompt={args.prompt_len} "
           f"generated={total_tokens} tokens (mixed 16..256)")
-    print(f"continuous: {t_cont:.2f} s = {total_tokens / t_cont:.0f} tok/s")
-    print(f"static    : {t_stat:.2f} s = {total_tokens / t_stat:.0f} tok/s "
-          f"(pays {sum(max(lens[j] for j in range(i, min(i + args.slots, args.requests)))  * min(args.slots, args.requests - i) for i in range(0, args.requests, args.slots)) - total_tokens} padded decode positions)")
-    print(f"speedup   : {t_stat / t_cont:.2f}x")
+    print(f"continuous: {t_cont:.2f} s = {total_tokens / t_cont:.0f} tok/s, "
+          f"mean completion latency {lat_c:.2f} s")
+    print(f"static    : {t_stat:.2f} s = {total_tokens / t_stat:.0f} tok/s, "
+          f"mean completion latency {lat_s:.2f} s "
+          f"(pays {padded} padded decode positions; short requests wait for "
+          f"their group)")
+    print(f"throughput ratio cont/static: {t_stat / t_cont:.2f}x; "
+          f"latency ratio static/cont: {lat_s / lat_c:.2f}x")
     return 0
 
 
