@@ -122,6 +122,8 @@ class ContinuousBatchingEngine:
         """One device->host sync per tick (per-slot .item() calls dominated
         the first version's decode time)."""
         logits = torch.stack([self._last_logits[s] for s in slots])  # [N, V]
+        if all(self.active[s].temperature == 0 for s in slots):
+            return logits.argmax(dim=-1).tolist()
         temps = torch.tensor([self.active[s].temperature for s in slots],
                              device=logits.device)
         toks = logits.argmax(dim=-1)
