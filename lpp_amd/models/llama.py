@@ -144,10 +144,14 @@ class GatherKVCache:
     per_row = True
 
     def __init__(self, parent: RaggedKVCache, slots: torch.Tensor,
-                 contiguous_range: Optional[tuple] = None):
+                 contiguous_range: Optional[tuple] = None,
+                 total_hint: Optional[int] = None):
         self.parent = parent
         self.slots = slots  # long tensor of active slot indices
         self.range = contiguous_range  # (lo, hi) half-open, or None
+        # host-known max row length AFTER this tick's append — avoids a
+        # device sync (lengths.max().item()) per layer per decode tick
+        self.total_hint = total_hint
 
     @property
     def k(self):
@@ -221,7 +225,8 @@ class LlamaAttention(nn.Module):
             q = ops.apply_rope_positions(q, cos, sin, lens)
             k = ops.apply_rope_positions(k, cos, sin, lens)
             new_lens = cache.append_one(k, v)
-            T = int(new_lens.max())
+            hint = getattr(cache, "total_hint", None)
+            T = hint if hint is not None else int(new_lens.max())
             if hasattr(cache, "kv_to"):
                 kc, vc = cache.kv_to(T)
             else:

@@ -177,6 +177,11 @@ class ContinuousBatchingEngine:
             self._views = [GatherKVCache(c, slot_t, contiguous_range=rng)
                            for c in self.caches]
             self._views_key = key
+        # host-side max length after this tick's scatter (no device sync)
+        total = max(self.active[s].prompt.numel() + len(self.active[s].generated)
+                    for s in slots) + 1
+        for v in self._views:
+            v.total_hint = total
         toks = torch.tensor([self.active[s].generated[-1] for s in slots],
                             dtype=torch.long, device=self.device).view(-1, 1)
         logits = self._run_layers(toks, self._views)[:, -1]
