@@ -221,9 +221,14 @@ class LlamaAttention(nn.Module):
             # sits at its OWN position; RoPE phases gathered per row and
             # attention masked by per-row key validity.
             assert S == 1, "ragged cache is a single-token decode contract"
-            lens = cache.lengths  # positions BEFORE append
-            q = ops.apply_rope_positions(q, cos, sin, lens)
-            k = ops.apply_rope_positions(k, cos, sin, lens)
+            tick = getattr(cache, "tick", None)  # per-tick hoisted tensors
+            if tick is not None:
+                q = ops.apply_rope_cs(q, tick.cos, tick.sin)
+                k = ops.apply_rope_cs(k, tick.cos, tick.sin)
+            else:
+                lens = cache.lengths  # positions BEFORE append
+                q = ops.apply_rope_positions(q, cos, sin, lens)
+                k = ops.apply_rope_positions(k, cos, sin, lens)
             new_lens = cache.append_one(k, v)
             hint = getattr(cache, "total_hint", None)
             T = hint if hint is not None else int(new_lens.max())
@@ -238,8 +243,11 @@ class LlamaAttention(nn.Module):
             if rep > 1:
                 kt = kt.repeat_interleave(rep, dim=1)
                 vt = vt.repeat_interleave(rep, dim=1)
-            mask = (torch.arange(T, device=x.device)[None, :]
-                    < new_lens[:, None])[:, None, None, :]
+            if tick is not None:
+                mask = tick.mask
+            else:
+                mask = (torch.arange(T, device=x.device)[None, :]
+                        < new_lens[:, None])[:, None, None, :]
             o = torch.nn.functional.scaled_dot_product_attention(
                 qt, kt, vt, attn_mask=mask
             ).transpose(1, 2).contiguous()

@@ -94,7 +94,8 @@ def use_hip(*tensors: torch.Tensor) -> bool:
 
 
 from .rmsnorm import rmsnorm  # noqa: E402
-from .rope import build_rope_cache, apply_rope, apply_rope_positions  # noqa: E402
+from .rope import (build_rope_cache, apply_rope, apply_rope_positions,  # noqa: E402
+                   apply_rope_cs)
 from .swiglu import swiglu  # noqa: E402
 from .cross_entropy import shifted_cross_entropy  # noqa: E402
 from .attention import causal_attention  # noqa: E402
@@ -104,6 +105,7 @@ __all__ = [
     "build_rope_cache",
     "apply_rope",
     "apply_rope_positions",
+    "apply_rope_cs",
     "swiglu",
     "shifted_cross_entropy",
     "causal_attention",

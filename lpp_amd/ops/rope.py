@@ -81,6 +81,17 @@ def apply_rope(
 
 
 @torch.no_grad()
+def apply_rope_cs(x: torch.Tensor, c: torch.Tensor, s: torch.Tensor) -> torch.Tensor:
+    """Apply RoPE with pre-gathered per-row cos/sin ([N,1,1,D/2]) — the
+    gather is identical for every layer of a decode tick, so the serving
+    engine hoists it (see apply_rope_positions)."""
+    D = x.shape[-1]
+    xf = x.float()
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
+    return torch.cat((x1 * c - x2 * s, x2 * c + x1 * s), dim=-1).to(x.dtype)
+
+
+@torch.no_grad()
 def apply_rope_positions(
     x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, positions: torch.Tensor
 ) -> torch.Tensor:
@@ -91,6 +102,4 @@ def apply_rope_positions(
     assert S == 1, "positions variant is a single-token decode contract"
     c = cos[positions].view(N, 1, 1, D // 2).to(torch.float32)
     s = sin[positions].view(N, 1, 1, D // 2).to(torch.float32)
-    xf = x.float()
-    x1, x2 = xf[..., : D // 2], xf[..., D // 2 :]
-    return torch.cat((x1 * c - x2 * s, x2 * c + x1 * s), dim=-1).to(x.dtype)
+    return apply_rope_cs(x, c, s)

@@ -177,11 +177,32 @@ class ContinuousBatchingEngine:
             self._views = [GatherKVCache(c, slot_t, contiguous_range=rng)
                            for c in self.caches]
             self._views_key = key
-        # host-side max length after this tick's scatter (no device sync)
-        total = max(self.active[s].prompt.numel() + len(self.active[s].generated)
-                    for s in slots) + 1
+        # host-side per-tick tensors, shared by EVERY layer: max length,
+        # gathered RoPE phases at each row's position, key-validity mask —
+        # rebuilding these per layer made the decode tick launch-bound
+        # the token being fed is generated[-1]: its position = tokens
+        # already in the cache = prompt + len(generated) - 1
+        lens_h = [self.active[s].prompt.numel() + len(self.active[s].generated) - 1
+                  for s in slots]
+        total = max(lens_h) + 1
+        from lpp_amd import ops
+
+        cfg = self.model.cfg
+        cos, sin = ops.build_rope_cache(cfg.max_seq_len, cfg.head_dim,
+                                        cfg.rope_theta, self.device)
+        pos = torch.tensor(lens_h, dtype=torch.long, device=self.device)
+
+        class _Tick:
+            pass
+
+        tick = _Tick()
+        tick.cos = cos[pos].view(len(slots), 1, 1, -1)
+        tick.sin = sin[pos].view(len(slots), 1, 1, -1)
+        tick.mask = (torch.arange(total, device=self.device)[None, :]
+                     <= pos[:, None])[:, None, None, :]
         for v in self._views:
             v.total_hint = total
+            v.tick = tick
         toks = torch.tensor([self.active[s].generated[-1] for s in slots],
                             dtype=torch.long, device=self.device).view(-1, 1)
         logits = self._run_layers(toks, self._views)[:, -1]
