@@ -84,3 +84,21 @@ def test_continuous_batching_rejects_oversized():
         raise AssertionError("expected ValueError")
     except ValueError:
         pass
+
+
+def test_continuous_batching_non_contiguous_slots():
+    """A middle slot retiring with an empty queue leaves holes — the decode
+    view must fall back to the gather (non-contiguous) path correctly."""
+    m, cfg = _model()
+    prompts = _prompts(cfg, seed=21)
+    # slot 1's request finishes first; no waiting request refills it
+    eng = ContinuousBatchingEngine(m, max_slots=3, max_seq_len=64)
+    eng.submit(Request("a", prompts[0], 8))
+    eng.submit(Request("b", prompts[1], 2))   # retires early -> hole at slot 1
+    eng.submit(Request("c", prompts[2], 8))
+    while eng.pending():
+        eng.step()
+    for uid, p, n in (("a", prompts[0], 8), ("b", prompts[1], 2),
+                      ("c", prompts[2], 8)):
+        ref = m.generate(p.view(1, -1), max_new_tokens=n)[0]
+        assert torch.equal(eng.results[uid], ref.cpu()), uid
