@@ -127,3 +127,45 @@ def test_pad_vocab(tmp_path):
     padded = convert2ckpt._pad_rows(sd["weight"], 300)
     assert padded.shape[0] == 300
     assert float(padded[256:].abs().sum()) == 0.0
+
+
+def test_hf_sharded_bins_and_safetensors(hf_tiny, tmp_path):
+    """A 65B convert streams sharded checkpoints; sharded .bin files and
+    .safetensors must produce the identical layer files as a single bin."""
+    hf_model, hf_dir = hf_tiny
+    sd = hf_model.state_dict()
+
+    # sharded bins
+    keys = sorted(sd)
+    shards = [keys[i::3] for i in range(3)]
+    bdir = tmp_path / "sharded"
+    bdir.mkdir()
+    for i, ks in enumerate(shards):
+        torch.save({k: sd[k] for k in ks},
+                   bdir / f"pytorch_model-{i + 1:05d}-of-00003.bin")
+    out_b = tmp_path / "out_b"
+    convert2ckpt.convert_hf(bdir, out_b, pad_vocab_to=0, dtype=None)
+
+    # safetensors
+    st = pytest.importorskip("safetensors.torch")
+    sdir = tmp_path / "safet"
+    sdir.mkdir()
+    st.save_file({k: v.contiguous() for k, v in sd.items()},
+                 str(sdir / "model.safetensors"))
+    out_s = tmp_path / "out_s"
+    convert2ckpt.convert_hf(sdir, out_s, pad_vocab_to=0, dtype=None)
+
+    # reference: single-bin convert
+    out_ref = tmp_path / "out_ref"
+    convert2ckpt.convert_hf(hf_dir, out_ref, pad_vocab_to=0, dtype=None)
+
+    for out in (out_b, out_s):
+        ref_files = sorted((out_ref / "global_step001").glob("layer_*.pt"))
+        got_files = sorted((out / "global_step001").glob("layer_*.pt"))
+        assert [f.name for f in got_files] == [f.name for f in ref_files]
+        for rf, gf in zip(ref_files, got_files):
+            a = torch.load(rf, weights_only=True)
+            b = torch.load(gf, weights_only=True)
+            assert sorted(a) == sorted(b), rf.name
+            for k in a:
+                assert torch.equal(a[k], b[k]), (rf.name, k)
